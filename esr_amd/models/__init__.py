@@ -1,0 +1,3 @@
+from .registry import register_model, build_model, get_model_cls, list_models  # noqa: F401
+from .esrnet import ESRNet  # noqa: F401
+from . import blocks  # noqa: F401

@@ -1,0 +1,315 @@
+"""ESRNet — the recurrent event-stream super-resolution network.
+
+Architecture parity with the reference's ``DeepRecurrNet``
+(ESR:models/model.py:294-344): head conv -> 3-stage stride-2 encoder ->
+temporal propagation (local triplet gating + bi-directional ConvGRU with
+persistent state across forward calls = the BPTT link) -> spatio-temporal
+fusion (deformable alignment to the centre frame + attention + 3-level
+decoder) -> tail conv, with /8 padding and output crop.
+
+MI355X-first deltas (math-preserving unless noted):
+  * the bi-directional GRU scan batches the forward and reverse direction
+    into one conv call per step (2x fewer kernel launches, same math);
+  * the decoder upsampler is config-selectable: 'bilinear' (reference
+    parity, interp+conv) or 'pixelshuffle' (sub-pixel conv; ~2.5x less HBM
+    traffic — the default for benchmarks; changes the parametrization, not
+    the interface);
+  * deformable alignment runs on the hand-written gfx950 HIP kernels
+    (esr_amd.ops.dcn).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops.dcn import DeformAlign2d
+from .blocks import (ConvLayer, MLP, RecurrentConvLayer, ResidualBlock,
+                     UpsampleConvLayer, PixelShuffleUpsample)
+from .registry import register_model
+
+__all__ = ["ESRNet"]
+
+
+def _pad_multiple(x: torch.Tensor, mult: int):
+    """Centre-pad the trailing H,W dims to a multiple of `mult`.
+
+    Parity: ESR:models/model_util.py:133-164 (CropSize pad/crop: ceil on
+    top/left).  Returns (padded, (top, left, H, W)).
+    """
+    H, W = x.shape[-2:]
+    Hc = math.ceil(H / mult) * mult
+    Wc = math.ceil(W / mult) * mult
+    if Hc == H and Wc == W:
+        return x, None
+    top = math.ceil((Hc - H) / 2)
+    bottom = Hc - H - top
+    left = math.ceil((Wc - W) / 2)
+    right = Wc - W - left
+    return F.pad(x, (left, right, top, bottom)), (top, left, H, W)
+
+
+def _crop(x: torch.Tensor, box):
+    if box is None:
+        return x
+    top, left, H, W = box
+    return x[..., top:top + H, left:left + W].contiguous()
+
+
+class FeatsExtract(nn.Module):
+    """3x stride-2 conv pyramid; returns features deepest-first
+    (parity: ESR:models/model.py:20-45)."""
+
+    def __init__(self, basech=16, norm=None, activation="relu"):
+        super().__init__()
+        self.convblock = nn.ModuleList([
+            ConvLayer(basech, 2 * basech, 3, 2, 1, activation, norm),
+            ConvLayer(2 * basech, 4 * basech, 3, 2, 1, activation, norm),
+            ConvLayer(4 * basech, 8 * basech, 3, 2, 1, activation, norm),
+        ])
+
+    def forward(self, x):
+        outs = []
+        for blk in self.convblock:
+            x = blk(x)
+            outs.append(x)
+        return outs[::-1]
+
+
+class TimePropagation(nn.Module):
+    """Local (triplet gating) + global (bi-dir ConvGRU) temporal correlation
+    with persistent hidden state (parity: ESR:models/model.py:48-153)."""
+
+    def __init__(self, basech=16, norm=None, activation="relu",
+                 has_ltc=True, has_gtc=True, gtc_frozen=False,
+                 recurrent_block_type="convgru"):
+        super().__init__()
+        assert has_ltc or has_gtc
+        self.has_ltc = has_ltc
+        self.has_gtc = has_gtc
+        self.gtc_frozen = gtc_frozen
+
+        if has_ltc:
+            self.pred_map = nn.Sequential(
+                ConvLayer(2 * basech, basech, 3, 1, 1, activation, norm),
+                ConvLayer(basech, 1, 3, 1, 1, "sigmoid", norm),
+            )
+            self.local_fusion = nn.Sequential(
+                ResidualBlock(3 * basech, 3 * basech, norm=norm),
+                ConvLayer(3 * basech, basech, 3, 1, 1, None, norm),
+            )
+        if has_gtc:
+            self.lstm = RecurrentConvLayer(basech, basech, 3, 1, 1,
+                                           recurrent_block_type, activation, norm)
+            self.global_fusion = ConvLayer(2 * basech, basech, 1, 1, 0,
+                                           activation, norm)
+        self.state = None  # concatenated (fwd|bwd) recurrent state
+
+    def reset_states(self):
+        self.state = None
+
+    def detach_states(self):
+        """Truncate BPTT: keep state values, drop the autograd history."""
+        def _d(s):
+            if s is None:
+                return None
+            if isinstance(s, tuple):
+                return tuple(_d(v) for v in s)
+            return s.detach()
+        self.state = _d(self.state)
+
+    def _local(self, f0, f1, f2):
+        m0 = self.pred_map(torch.cat([f0, f1], dim=1))
+        m1 = self.pred_map(torch.cat([f1, f2], dim=1))
+        out = self.local_fusion(torch.cat([f0 * m0, f1, f2 * m1], dim=1))
+        return out + f1
+
+    def _global(self, feats):
+        B, N, C, H, W = feats.shape
+        rev_idx = list(reversed(range(N)))
+        rev = feats[:, rev_idx]
+        state = None if self.gtc_frozen else self.state
+        outs = []
+        for i in range(N):
+            # fwd and bwd scan share weights -> run both in one batched call
+            step_in = torch.cat([feats[:, i], rev[:, i]], dim=0)     # [2B,...]
+            out, state = self.lstm(step_in, state)
+            if self.gtc_frozen:
+                state = None
+            outs.append(out)
+        self.state = None if self.gtc_frozen else state
+
+        both = torch.stack(outs, dim=1)                               # [2B,N,...]
+        x = both[:B]
+        r = both[B:][:, rev_idx]
+        fused = torch.cat([x, r], dim=2).view(B * N, 2 * C, H, W)
+        fused = self.global_fusion(fused)
+        return fused.view(B, N, C, H, W)
+
+    def forward(self, x):
+        B, N, C, H, W = x.shape
+        if self.has_ltc:
+            feats = []
+            for i in range(N):
+                i0, i1, i2 = (0, 0, 1) if i == 0 else \
+                             ((N - 2, N - 1, N - 1) if i == N - 1 else (i - 1, i, i + 1))
+                feats.append(self._local(x[:, i0], x[:, i1], x[:, i2]))
+            feats = torch.stack(feats, dim=1)
+        else:
+            feats = x
+        if self.has_gtc:
+            feats = self._global(feats)
+        return feats + x
+
+
+class STFusion(nn.Module):
+    """Deformable alignment of each frame to the centre frame, spatial +
+    channel attention fusion, and 3-level decoder
+    (parity: ESR:models/model.py:156-291)."""
+
+    def __init__(self, basech=16, num_frame=3, norm=None, activation="relu",
+                 has_dcnatten=True, has_scaleaggre=True, upsampler="bilinear",
+                 deformable_groups=8):
+        super().__init__()
+        assert has_dcnatten or has_scaleaggre
+        assert num_frame >= 3 and (num_frame + 1) % 2 == 0
+        self.has_dcnatten = has_dcnatten
+        self.has_scaleaggre = has_scaleaggre
+        self.num_frame = num_frame
+        self.mid_idx = (num_frame - 1) // 2
+
+        if has_dcnatten:
+            self.offset = nn.Sequential(
+                ConvLayer(2 * basech, basech, 3, 1, 1, activation, norm),
+                ConvLayer(basech, basech, 3, 1, 1, None, norm),
+            )
+            self.dcn = DeformAlign2d(basech, basech, 3, stride=1, padding=1,
+                                     dilation=1, deformable_groups=deformable_groups)
+            self.convblock = nn.Sequential(
+                ConvLayer(2 * basech, basech, 3, 1, 1, activation, norm),
+                ConvLayer(basech, basech, 3, 1, 1, None, norm),
+            )
+            self.kernel = ConvLayer(basech, 2, 1, 1, 0, "sigmoid", norm)
+            self.fc = nn.Sequential(
+                MLP(basech, basech // 2, 2 * basech, 2), nn.Sigmoid())
+            self.dcn_fusion = nn.Sequential(
+                ConvLayer(2 * basech, basech, 3, 1, 1, activation, norm),
+                ConvLayer(basech, basech, 3, 1, 1, None, norm),
+            )
+
+        self.dense_fusion = nn.Sequential(
+            ConvLayer(num_frame * basech, basech, 3, 1, 1, activation, norm),
+            ConvLayer(basech, basech, 3, 1, 1, None, norm),
+        )
+
+        if has_scaleaggre:
+            self.attens = nn.ModuleList([
+                ConvLayer(basech, 1, 3, 1, 1, "sigmoid", norm),
+                ConvLayer(basech // 2, 1, 3, 1, 1, "sigmoid", norm),
+                ConvLayer(basech // 4, 1, 3, 1, 1, "sigmoid", norm),
+            ])
+
+        Up = {"bilinear": UpsampleConvLayer,
+              "pixelshuffle": PixelShuffleUpsample}[upsampler]
+        self.recons = nn.ModuleList([
+            Up(basech, basech // 2, 3, 1, 1, norm=norm),
+            Up(basech // 2, basech // 4, 3, 1, 1, norm=norm),
+            Up(basech // 4, basech // 8, 3, 1, 1, norm=norm),
+        ])
+
+    def fuse(self, feat0, feat1):
+        B, C, H, W = feat0.shape
+        offset_feat = self.offset(torch.cat([feat0, feat1], dim=1))
+        aligned = F.relu(self.dcn(feat0, offset_feat))
+        feat = self.convblock(torch.cat([aligned, feat1], dim=1))
+        spatial_k = self.kernel(feat)                                  # [B,2,H,W]
+        pooled = feat.view(B, C, H * W).transpose(1, 2).max(1, keepdim=True)[0]
+        channel_k = self.fc(pooled).transpose(1, 2).unsqueeze(-1)      # [B,2C,1,1]
+        y0 = aligned * spatial_k[:, :1] * channel_k[:, :C]
+        y1 = feat1 * spatial_k[:, 1:2] * channel_k[:, C:]
+        return self.dcn_fusion(torch.cat([y0, y1], dim=1))
+
+    def dense_fuse(self, x):
+        if self.has_dcnatten:
+            mid = x[:, self.mid_idx].contiguous()
+            outs = [self.fuse(x[:, i].contiguous(), mid)
+                    for i in range(self.num_frame) if i != self.mid_idx]
+            outs.append(mid)
+            out = torch.cat(outs, dim=1)
+        else:
+            out = x.reshape(x.size(0), -1, x.size(-2), x.size(-1))
+        return self.dense_fusion(out)
+
+    def scale_aggre(self, x, feats, idx):
+        if self.has_scaleaggre:
+            B, N, C, H, W = feats.shape
+            flat = feats.reshape(B * N, C, H, W)
+            flat = flat * self.attens[idx](flat)
+            x = x + flat.view(B, N, C, H, W).mean(1)
+        return self.recons[idx](x)
+
+    def forward(self, x, feats_list):
+        B, N = x.shape[:2]
+        assert N == self.num_frame
+        out = self.dense_fuse(x)
+        for idx, feats in enumerate(feats_list):
+            feats = feats.view(B, N, -1, feats.size(-2), feats.size(-1))
+            out = self.scale_aggre(out, feats, idx)
+        return out
+
+
+@register_model("ESRNet")
+@register_model("DeepRecurrNet")  # reference-config compatibility alias
+class ESRNet(nn.Module):
+    """Recurrent event-stream SR network (reference: DeepRecurrNet,
+    ESR:models/model.py:294-344).
+
+    Input: [B, N, inch, kH, kW] scaled count maps (LR events splatted on the
+    HR grid); output: [B, inch, kH, kW] predicted HR count map for the
+    middle frame.
+    """
+
+    DOWN_SCALE = 8
+
+    def __init__(self, inch=2, basech=16, num_frame=3, norm=None,
+                 activation="relu", has_ltc=True, has_gtc=True,
+                 gtc_frozen=False, has_dcnatten=True, has_scaleaggre=True,
+                 upsampler="bilinear", recurrent_block_type="convgru",
+                 deformable_groups=8):
+        super().__init__()
+        d = self.DOWN_SCALE
+        self.head = ConvLayer(inch, basech, 3, 1, 1, activation, norm)
+        self.feat_extract = FeatsExtract(basech, norm, activation)
+        self.time_propagate = TimePropagation(
+            d * basech, norm, activation, has_ltc, has_gtc, gtc_frozen,
+            recurrent_block_type)
+        self.spacetime_fuse = STFusion(
+            d * basech, num_frame, norm, activation, has_dcnatten,
+            has_scaleaggre, upsampler, deformable_groups)
+        self.tail = ConvLayer(basech, inch, 3, 1, 1, "relu", norm)
+
+    def reset_states(self):
+        self.time_propagate.reset_states()
+
+    def detach_states(self):
+        self.time_propagate.detach_states()
+
+    @property
+    def num_parameters(self):
+        return sum(p.numel() for p in self.parameters() if p.requires_grad)
+
+    def forward(self, x):
+        B, N, C, H, W = x.shape
+        x, box = _pad_multiple(x, self.DOWN_SCALE)
+        x = x.reshape(B * N, C, x.size(-2), x.size(-1))
+        x = self.head(x)
+        feats_list = self.feat_extract(x)
+        deep = feats_list[0]
+        deep = deep.view(B, N, -1, deep.size(-2), deep.size(-1))
+        deep = self.time_propagate(deep)
+        out = self.spacetime_fuse(deep, feats_list)
+        out = self.tail(out)
+        return _crop(out, box)

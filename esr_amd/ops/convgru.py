@@ -1,0 +1,114 @@
+"""ConvGRU cell with fused gate math.
+
+The reference computes the GRU gate math as ~8 separate elementwise CUDA ops
+around three independent convolutions (ESR:models/submodules.py:474-514).
+MI355X-first redesign:
+
+  * update+reset gates come from ONE conv (2*hidden channels, single MIOpen
+    call) instead of two;
+  * all elementwise gate math is fused into two HIP kernels
+    (gru_gates_ur: sigmoid(u), sigmoid(r), h*r  /  gru_gates_out:
+    h_new = h*(1-u) + tanh(o)*u), each with a hand-written backward —
+    HBM-bound work is read once, written once instead of round-tripping per
+    op (guide: fuse elementwise work, G13 vectorized bf16 loads).
+
+CPU fallback (and autograd oracle) is plain torch ops with identical math.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from .native import get_ext, require_ext
+
+__all__ = ["ConvGRUCell", "gru_gates_ur", "gru_gates_out"]
+
+
+class _GatesUR(torch.autograd.Function):
+    """(ur_pre [B,2C,H,W], h [B,C,H,W]) -> (u, r, h*r) fused."""
+
+    @staticmethod
+    def forward(ctx, ur_pre, h):
+        ext = require_ext()
+        u, r, hr = ext.gru_gates_ur_forward(ur_pre, h)
+        ctx.save_for_backward(u, r, h)
+        return u, r, hr
+
+    @staticmethod
+    def backward(ctx, du, dr, dhr):
+        u, r, h = ctx.saved_tensors
+        ext = require_ext()
+        d_ur, dh = ext.gru_gates_ur_backward(
+            du.contiguous(), dr.contiguous(), dhr.contiguous(), u, r, h)
+        return d_ur, dh
+
+
+class _GatesOut(torch.autograd.Function):
+    """(o_pre, u, h) -> h_new = h*(1-u) + tanh(o_pre)*u fused."""
+
+    @staticmethod
+    def forward(ctx, o_pre, u, h):
+        ext = require_ext()
+        h_new, tanh_o = ext.gru_gates_out_forward(o_pre, u, h)
+        ctx.save_for_backward(u, h, tanh_o)
+        return h_new
+
+    @staticmethod
+    def backward(ctx, dh_new):
+        u, h, tanh_o = ctx.saved_tensors
+        ext = require_ext()
+        do_pre, du, dh = ext.gru_gates_out_backward(dh_new.contiguous(), u, h, tanh_o)
+        return do_pre, du, dh
+
+
+def gru_gates_ur(ur_pre: torch.Tensor, h: torch.Tensor):
+    if ur_pre.is_cuda and get_ext() is not None and ur_pre.dtype == torch.float32:
+        return _GatesUR.apply(ur_pre, h)
+    C = h.size(1)
+    u = torch.sigmoid(ur_pre[:, :C])
+    r = torch.sigmoid(ur_pre[:, C:])
+    return u, r, h * r
+
+
+def gru_gates_out(o_pre: torch.Tensor, u: torch.Tensor, h: torch.Tensor):
+    if o_pre.is_cuda and get_ext() is not None and o_pre.dtype == torch.float32:
+        return _GatesOut.apply(o_pre, u, h)
+    return h * (1 - u) + torch.tanh(o_pre) * u
+
+
+class ConvGRUCell(nn.Module):
+    """Convolutional GRU cell (math parity: ESR:models/submodules.py:474-514).
+
+    h_new = h*(1-u) + tanh(conv_o([x, h*r]))*u,
+    u = sigmoid(conv_u([x,h])), r = sigmoid(conv_r([x,h])).
+
+    conv_u and conv_r are fused into a single 2C-output convolution; weights
+    are orthogonally initialized per gate like the reference.
+    """
+
+    def __init__(self, input_size: int, hidden_size: int, kernel_size: int = 3):
+        super().__init__()
+        padding = kernel_size // 2
+        self.input_size = input_size
+        self.hidden_size = hidden_size
+        self.ur_gate = nn.Conv2d(input_size + hidden_size, 2 * hidden_size,
+                                 kernel_size, padding=padding)
+        self.out_gate = nn.Conv2d(input_size + hidden_size, hidden_size,
+                                  kernel_size, padding=padding)
+        # orthogonal init per gate (ESR:models/submodules.py:489-494)
+        nn.init.orthogonal_(self.ur_gate.weight[: hidden_size])
+        nn.init.orthogonal_(self.ur_gate.weight[hidden_size:])
+        nn.init.orthogonal_(self.out_gate.weight)
+        nn.init.zeros_(self.ur_gate.bias)
+        nn.init.zeros_(self.out_gate.bias)
+
+    def forward(self, x: torch.Tensor, h: torch.Tensor | None):
+        if h is None:
+            h = torch.zeros(x.size(0), self.hidden_size, x.size(2), x.size(3),
+                            dtype=x.dtype, device=x.device)
+        xh = torch.cat([x, h], dim=1)
+        ur_pre = self.ur_gate(xh)
+        u, r, hr = gru_gates_ur(ur_pre, h)
+        o_pre = self.out_gate(torch.cat([x, hr], dim=1))
+        return gru_gates_out(o_pre, u, h)

@@ -1,0 +1,3 @@
+from .ddp import (init_distributed, is_distributed, get_rank, get_world_size,  # noqa: F401
+                  barrier, wrap_ddp, setup_rank0_print)
+from .reduce import reduce_tensor, reduce_dict  # noqa: F401

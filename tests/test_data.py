@@ -1,0 +1,158 @@
+"""Data-layer tests: EVS store, scale pairs, windowing, sequences, collate."""
+
+import numpy as np
+import pytest
+import torch
+
+from esr_amd.data import (EventSRDataset, EventStore, SequenceDataset,
+                          SequenceDataLoader, read_datalist,
+                          resolve_scale_pair, write_synthetic_store)
+
+
+def _ds_config(**over):
+    cfg = {
+        "scale": 2, "ori_scale": "down4", "time_bins": 1,
+        "need_gt_frame": True, "need_gt_events": True,
+        "mode": "events", "window": 512, "sliding_window": 256,
+        "data_augment": {"enabled": False, "augment": [], "augment_prob": []},
+        "hot_filter": {"enabled": False},
+        "sequence": {"sequence_length": 4, "seqn": 3, "step_size": None,
+                     "pause": {"enabled": False,
+                               "proba_pause_when_running": 0.05,
+                               "proba_pause_when_paused": 0.9}},
+    }
+    cfg.update(over)
+    return cfg
+
+
+@pytest.fixture(scope="module")
+def store_path(tmp_path_factory):
+    p = tmp_path_factory.mktemp("evs") / "seq.evs"
+    write_synthetic_store(p, resolution=(64, 64), num_events=80_000, seed=1)
+    return p
+
+
+def test_store_roundtrip(store_path):
+    s = EventStore(store_path)
+    assert s.sensor_resolution == [64, 64]
+    assert set(s.groups) == {"ori", "down2", "down4", "down8", "down16"}
+    ev = s.events("ori", 0, 100)
+    assert ev.shape == (4, 100)
+    assert (np.diff(s.ts("ori")[:1000]) >= 0).all()
+    # thinning: down2 has ~1/4 the events
+    assert abs(s.num_events("down2") * 4 - s.num_events("ori")) < 8
+
+
+def test_resolve_scale_pair_matrix():
+    # (ori_scale, scale) -> (inp_prefix, gt_prefix) with gt events
+    cases = {
+        ("down16", 2): ("down16", "down8"),
+        ("down16", 4): ("down16", "down4"),
+        ("down8", 2): ("down8", "down4"),
+        ("down4", 4): ("down4", "ori"),
+        ("down2", 2): ("down2", "ori"),
+    }
+    for (os_, sc), (ip, gp) in cases.items():
+        inp_p, inp_r, gt_p, gt_r, down_r = resolve_scale_pair(
+            (256, 256), os_, sc, need_gt_events=True)
+        assert inp_p == ip and gt_p == gp
+        assert gt_r[0] == inp_r[0] * sc
+    # no GT events: synthetic upscale target
+    _, inp_r, gt_p, gt_r, _ = resolve_scale_pair((256, 256), "down4", 2,
+                                                 need_gt_events=False)
+    assert gt_p == "down4" and gt_r == [r * 2 for r in inp_r]
+    with pytest.raises(ValueError):
+        resolve_scale_pair((256, 256), "down2", 4, need_gt_events=True)
+
+
+def test_dataset_item(store_path):
+    ds = EventSRDataset(store_path, _ds_config())
+    assert len(ds) > 0
+    item = ds[0]
+    H, W = ds.inp_sensor_resolution
+    kH, kW = ds.gt_sensor_resolution
+    assert kH == 2 * H
+    assert item["inp_cnt"].shape == (2, H, W)
+    assert item["inp_scaled_cnt"].shape == (2, kH, kW)
+    assert item["gt_cnt"].shape == (2, kH, kW)
+    assert item["inp_stack"].shape == (1, H, W)
+    assert item["gt_img"].shape == (1, kH, kW)
+    # count conservation: total inp counts == window size (all in range)
+    assert item["inp_cnt"].sum() <= 512
+    assert item["inp_scaled_cnt"].sum() == item["inp_cnt"].sum()
+
+
+def test_gt_alignment_count(store_path):
+    ds = EventSRDataset(store_path, _ds_config())
+    g0, g1 = ds.gt_event_indices[0]
+    i0, i1 = ds.event_indices[0]
+    assert (g1 - g0) == ds.scale ** 2 * (i1 - i0)
+
+
+def test_augmentation_determinism(store_path):
+    cfg = _ds_config()
+    cfg["data_augment"] = {"enabled": True,
+                           "augment": ["Horizontal", "Vertical", "Polarity"],
+                           "augment_prob": [1.0, 1.0, 1.0]}
+    ds = EventSRDataset(store_path, cfg)
+    a = ds.__getitem__(0, seed=42)
+    b = ds.__getitem__(0, seed=42)
+    assert torch.allclose(a["inp_cnt"], b["inp_cnt"])
+    ds_plain = EventSRDataset(store_path, _ds_config())
+    c = ds_plain.__getitem__(0, seed=42)
+    # full flips + polarity: pos/neg channels swapped and mirrored
+    assert torch.allclose(a["inp_cnt"][0], torch.flip(c["inp_cnt"][1], (0, 1)))
+
+
+def test_pause_zeroes_input(store_path):
+    ds = EventSRDataset(store_path, _ds_config())
+    item = ds.__getitem__(0, pause=True, seed=1)
+    assert item["inp_cnt"].sum() == 0
+    assert item["gt_cnt"].sum() > 0
+
+
+def test_noise_injection(store_path):
+    cfg = _ds_config()
+    cfg["add_noise"] = {"enabled": True, "noise_level": 0.05}
+    ds = EventSRDataset(store_path, cfg)
+    item = ds.__getitem__(0, seed=3)
+    base = EventSRDataset(store_path, _ds_config()).__getitem__(0, seed=3)
+    assert item["inp_cnt"].sum() > base["inp_cnt"].sum()
+
+
+def test_sequence_dataset(store_path):
+    sd = SequenceDataset(store_path, _ds_config())
+    assert len(sd) >= 1
+    seq = sd[0]
+    assert len(seq) == 4
+    assert all("inp_scaled_cnt" in it for it in seq)
+
+
+def test_sequence_loader_collate(synth_datalist):
+    cfg = {
+        "use_ddp": False, "path_to_datalist_txt": synth_datalist,
+        "batch_size": 2, "shuffle": False, "num_workers": 0,
+        "pin_memory": False, "drop_last": True,
+        "dataset": _ds_config(ori_scale="down4",
+                              window=1024, sliding_window=512),
+    }
+    loader = SequenceDataLoader(cfg)
+    batch = next(iter(loader))
+    # seql=4, seqn=3 -> 2 sliding windows
+    assert isinstance(batch, list) and len(batch) == 2
+    w = batch[0]
+    assert w["inp_scaled_cnt"].dim() == 5
+    assert w["inp_scaled_cnt"].shape[:2] == (2, 3)
+
+
+def test_time_mode_windowing(store_path):
+    cfg = _ds_config(mode="time", window=0.2, sliding_window=0.1)
+    ds = EventSRDataset(store_path, cfg)
+    assert len(ds) >= 1
+    item = ds[0]
+    assert item["inp_cnt"].sum() > 0
+
+
+def test_datalist(synth_datalist):
+    paths = read_datalist(synth_datalist)
+    assert len(paths) == 2

@@ -1,0 +1,83 @@
+"""ESRNet model-family tests."""
+
+import torch
+
+from esr_amd.models import build_model, list_models
+from esr_amd.ops.convgru import ConvGRUCell
+
+
+def test_registry():
+    assert "ESRNet" in list_models()
+    assert "DeepRecurrNet" in list_models()   # reference-config alias
+
+
+def test_forward_shapes_and_padcrop():
+    m = build_model("ESRNet", inch=2, basech=8, num_frame=3)
+    for H, W in [(32, 32), (36, 44), (40, 40)]:
+        y = m(torch.randn(1, 3, 2, H, W))
+        assert y.shape == (1, 2, H, W)
+        m.reset_states()
+
+
+def test_bptt_state_persistence_and_reset():
+    m = build_model("ESRNet", inch=2, basech=8, num_frame=3)
+    x = torch.randn(2, 3, 2, 32, 32)
+    m.reset_states()
+    y1 = m(x)
+    assert m.time_propagate.state is not None
+    y2 = m(x)           # state carried -> different output
+    assert not torch.allclose(y1, y2)
+    m.reset_states()
+    y3 = m(x)
+    assert torch.allclose(y1, y3, atol=1e-6)
+
+
+def test_bptt_backward_through_sequence():
+    m = build_model("ESRNet", inch=2, basech=8, num_frame=3)
+    m.reset_states()
+    loss = 0
+    for _ in range(3):
+        loss = loss + (m(torch.randn(1, 3, 2, 32, 32)) ** 2).mean()
+    loss.backward()
+    grads = [p.grad for p in m.parameters() if p.grad is not None]
+    assert len(grads) > 0
+    assert all(torch.isfinite(g).all() for g in grads)
+
+
+def test_convgru_cell_math():
+    torch.manual_seed(0)
+    cell = ConvGRUCell(4, 4, 3)
+    x = torch.randn(2, 4, 8, 8)
+    h = cell(x, None)
+    assert h.shape == (2, 4, 8, 8)
+    h2 = cell(x, h)
+    assert not torch.allclose(h, h2)
+    # manual recompute of the gate math
+    import torch.nn.functional as F
+    xh = torch.cat([x, h], 1)
+    ur = cell.ur_gate(xh)
+    u, r = torch.sigmoid(ur[:, :4]), torch.sigmoid(ur[:, 4:])
+    o = torch.tanh(cell.out_gate(torch.cat([x, h * r], 1)))
+    ref = h * (1 - u) + o * u
+    assert torch.allclose(h2, ref, atol=1e-6)
+
+
+def test_pixelshuffle_upsampler_variant():
+    m = build_model("ESRNet", inch=2, basech=8, num_frame=3,
+                    upsampler="pixelshuffle")
+    y = m(torch.randn(1, 3, 2, 32, 32))
+    assert y.shape == (1, 2, 32, 32)
+
+
+def test_ablation_flags():
+    for kwargs in [dict(has_ltc=False), dict(has_gtc=False),
+                   dict(has_dcnatten=False), dict(has_scaleaggre=False)]:
+        m = build_model("ESRNet", inch=2, basech=8, num_frame=3, **kwargs)
+        y = m(torch.randn(1, 3, 2, 16, 16))
+        assert y.shape == (1, 2, 16, 16)
+
+
+def test_num_frame_5():
+    m = build_model("ESRNet", inch=2, basech=8, num_frame=5)
+    y = m(torch.randn(1, 5, 2, 16, 16))
+    assert y.shape == (1, 2, 16, 16)
