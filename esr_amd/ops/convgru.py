@@ -62,9 +62,13 @@ class _GatesOut(torch.autograd.Function):
         return do_pre, du, dh
 
 
+_FUSED_DTYPES = (torch.float32, torch.bfloat16)
+
+
 def gru_gates_ur(ur_pre: torch.Tensor, h: torch.Tensor):
-    if ur_pre.is_cuda and get_ext() is not None and ur_pre.dtype == torch.float32:
-        return _GatesUR.apply(ur_pre, h)
+    if ur_pre.is_cuda and get_ext() is not None \
+            and ur_pre.dtype in _FUSED_DTYPES and h.dtype == ur_pre.dtype:
+        return _GatesUR.apply(ur_pre.contiguous(), h.contiguous())
     C = h.size(1)
     u = torch.sigmoid(ur_pre[:, :C])
     r = torch.sigmoid(ur_pre[:, C:])
@@ -72,8 +76,11 @@ def gru_gates_ur(ur_pre: torch.Tensor, h: torch.Tensor):
 
 
 def gru_gates_out(o_pre: torch.Tensor, u: torch.Tensor, h: torch.Tensor):
-    if o_pre.is_cuda and get_ext() is not None and o_pre.dtype == torch.float32:
-        return _GatesOut.apply(o_pre, u, h)
+    if o_pre.is_cuda and get_ext() is not None \
+            and o_pre.dtype in _FUSED_DTYPES and u.dtype == o_pre.dtype \
+            and h.dtype == o_pre.dtype:
+        return _GatesOut.apply(o_pre.contiguous(), u.contiguous(),
+                               h.contiguous())
     return h * (1 - u) + torch.tanh(o_pre) * u
 
 
