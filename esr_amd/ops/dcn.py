@@ -100,8 +100,9 @@ class _DeformConvHIP(torch.autograd.Function):
     def forward(ctx, input, offset, mask, weight, bias,
                 stride, padding, dilation, deformable_groups):
         ext = get_ext()
-        out, columns = ext.deform_conv2d_forward(
-            input, offset, mask, weight, bias,
+        out = ext.deform_conv2d_forward(
+            input.contiguous(), offset.contiguous(), mask.contiguous(),
+            weight.contiguous(), bias,
             stride[0], stride[1], padding[0], padding[1],
             dilation[0], dilation[1], deformable_groups)
         ctx.save_for_backward(input, offset, mask, weight)
