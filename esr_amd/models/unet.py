@@ -8,7 +8,7 @@ from __future__ import annotations
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
+
 
 from .blocks import (ConvLayer, PixelShuffleUpsample, RecurrentConvLayer,
                      ResidualBlock, TransposedConvLayer, UpsampleConvLayer)
