@@ -3,9 +3,16 @@
 
 Measures the BASELINE.json headline metric — SR event-frames/sec for the
 whole job — on the named config: 2x ESR, NFS-syn-shaped synthetic data
-(window=2048 events, LR 128x128 -> HR 256x256), seq_len=8 / seqn=3 BPTT,
-bf16 autocast, Adam; one full BPTT optimizer step per iteration exactly as
+(window=2048 events per LR frame, LR 128x128 -> HR 256x256), seq_len=8 /
+seqn=3 BPTT, Adam; one full BPTT optimizer step per iteration exactly as
 the trainer does (ESR:train_ours_cnt_seq.py:210-235 semantics).
+
+MI355X design: the model is tiny (1.8 M params, ~130 kernels per window),
+so a step is launch-latency-bound — the whole BPTT step (6 windows of
+forward + one backward + Adam) is captured in a hipGraph and replayed;
+fresh synthetic data is copied into the graph's static input buffers every
+step.  Synthetic events are generated and splatted ON DEVICE with the
+native splat kernels.  --no-graphs falls back to eager.
 
 Single GPU:   python bench.py --gpus 1 --steps 20 --warmup 5
 Multi-GPU (driver-launched):
@@ -23,40 +30,42 @@ import torch.distributed as dist
 import torch.nn.functional as F
 
 
-def synth_sequences(n_seq, seql, seqn, batch, window, lr_res, hr_res, device,
-                    seed=0):
-    """Pre-generate GPU-resident (inp_scaled_cnt, gt_cnt) window sequences of
-    the benchmark shape using the native splat kernels."""
-    from esr_amd.data.synthetic import generate_events
-    from esr_amd.ops import events_to_channels, normalize_events, \
-        scaled_count_encoding
-    import numpy as np
-
-    seqs = []
-    rng = np.random.default_rng(seed)
-    n_windows = seql - seqn + 1
-    for s in range(n_seq):
+def make_window_pool(n_sets, n_windows, batch, seqn, window, lr_res, hr_res,
+                     device, ext, seed):
+    """Generate pools of (inp_scaled_cnt [B,seqn,2,kH,kW], gt_cnt [B,2,kH,kW])
+    per window, fully on device via the native splat kernels."""
+    g = torch.Generator(device=device).manual_seed(seed)
+    H, W = lr_res
+    kH, kW = hr_res
+    scale = kH // H
+    pools = []
+    for _ in range(n_sets):
         windows = []
-        for w in range(n_windows):
-            inp_frames, gt_frames = [], []
-            for b in range(batch):
-                for f in range(seqn):
-                    ev = generate_events(window, lr_res,
-                                         seed=int(rng.integers(1 << 30)))
-                    ev_t = torch.from_numpy(ev).float()
-                    norm = normalize_events(ev_t, lr_res)
-                    inp_frames.append(scaled_count_encoding(norm, hr_res, "cnt"))
-                    gt_ev = generate_events(window * 4, hr_res,
-                                            seed=int(rng.integers(1 << 30)))
-                    gt_t = torch.from_numpy(gt_ev).float()
-                    gt_frames.append(events_to_channels(gt_t[0], gt_t[1],
-                                                        gt_t[3], hr_res))
-            inp = torch.stack(inp_frames).view(batch, seqn, 2, *hr_res)
-            gt = torch.stack(gt_frames).view(batch, seqn, 2, *hr_res)
-            mid = (seqn - 1) // 2
-            windows.append((inp.to(device), gt[:, mid].contiguous().to(device)))
-        seqs.append(windows)
-    return seqs
+        for _ in range(n_windows):
+            BF = batch * seqn
+            ev = torch.empty(BF, window, 4, device=device)
+            ev[..., 0] = (torch.rand(BF, window, device=device, generator=g)
+                          * W).floor() * scale
+            ev[..., 1] = (torch.rand(BF, window, device=device, generator=g)
+                          * H).floor() * scale
+            ev[..., 2] = torch.rand(BF, window, device=device, generator=g)
+            ev[..., 3] = torch.randint(0, 2, (BF, window), device=device,
+                                       generator=g).float() * 2 - 1
+            inp = ext.splat_count(ev.contiguous(), kH, kW) \
+                .view(batch, seqn, 2, kH, kW)
+            gt_ev = torch.empty(batch, window * scale * scale, 4, device=device)
+            gt_ev[..., 0] = (torch.rand(batch, window * scale * scale,
+                                        device=device, generator=g) * kW).floor()
+            gt_ev[..., 1] = (torch.rand(batch, window * scale * scale,
+                                        device=device, generator=g) * kH).floor()
+            gt_ev[..., 2] = torch.rand(batch, window * scale * scale,
+                                       device=device, generator=g)
+            gt_ev[..., 3] = torch.randint(0, 2, (batch, window * scale * scale),
+                                          device=device, generator=g).float() * 2 - 1
+            gt = ext.splat_count(gt_ev.contiguous(), kH, kW)
+            windows.append((inp, gt))
+        pools.append(windows)
+    return pools
 
 
 def main():
@@ -74,6 +83,8 @@ def main():
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp32"])
     p.add_argument("--upsampler", type=str, default="pixelshuffle")
+    p.add_argument("--no-graphs", action="store_true",
+                   help="disable hipGraph capture (eager mode)")
     args = p.parse_args()
 
     assert torch.cuda.is_available(), "bench.py requires an MI355X GPU"
@@ -89,42 +100,121 @@ def main():
 
     from esr_amd.models import build_model
     from esr_amd.ops.native import require_ext
-    require_ext()  # fail loudly if HIP kernels are missing
+    ext = require_ext()  # fail loudly if HIP kernels are missing
 
     model = build_model("ESRNet", inch=2, basech=args.basech,
                         num_frame=args.seqn,
                         upsampler=args.upsampler).to(device)
-    if world > 1:
-        model = torch.nn.parallel.DistributedDataParallel(
-            model, device_ids=[local_rank], bucket_cap_mb=64,
-            gradient_as_bucket_view=True)
-    inner = model.module if hasattr(model, "module") else model
-    optimizer = torch.optim.Adam(model.parameters(), lr=1e-3,
-                                 weight_decay=1e-4, amsgrad=True)
+
+    # one flat fp32 gradient buffer -> a single RCCL all-reduce per step
+    params = [prm for prm in model.parameters() if prm.requires_grad]
+    total = sum(prm.numel() for prm in params)
+    flat_grad = torch.zeros(total, device=device)
+    off = 0
+    for prm in params:
+        prm.grad = flat_grad[off:off + prm.numel()].view_as(prm)
+        off += prm.numel()
+
+    optimizer = torch.optim.Adam(params, lr=1e-3, weight_decay=1e-4,
+                                 amsgrad=True, foreach=True,
+                                 capturable=not args.no_graphs)
 
     lr_res = (args.lr_size, args.lr_size)
     hr_res = (args.lr_size * args.scale, args.lr_size * args.scale)
-    seqs = synth_sequences(2, args.seql, args.seqn, args.batch, args.window,
-                           lr_res, hr_res, device, seed=100 + rank)
     n_windows = args.seql - args.seqn + 1
+    pools = make_window_pool(2, n_windows, args.batch, args.seqn, args.window,
+                             lr_res, hr_res, device, ext, seed=100 + rank)
     amp_dtype = torch.bfloat16 if args.dtype == "bf16" else None
 
-    def step(i):
-        windows = seqs[i % len(seqs)]
-        optimizer.zero_grad(set_to_none=True)
-        inner.reset_states()
+    # static input buffers (graph replay reads from fixed addresses)
+    static_in = [torch.empty_like(inp) for inp, _ in pools[0]]
+    static_gt = [torch.empty_like(gt) for _, gt in pools[0]]
+
+    import contextlib
+
+    def autocast():
+        if amp_dtype is None:
+            return contextlib.nullcontext()
+        # cache_enabled=False: the autocast weight-cast cache is not
+        # graph-capture-safe (casts must be recorded into the graph)
+        return torch.autocast("cuda", dtype=amp_dtype, cache_enabled=False)
+
+    def fwd_bwd():
+        flat_grad.zero_()
+        model.reset_states()
         loss = 0
-        for inp, gt in windows:
-            if amp_dtype is not None:
-                with torch.autocast("cuda", dtype=amp_dtype):
-                    pred = model(inp)
-                loss = loss + F.mse_loss(pred.float(), gt)
-            else:
+        for inp, gt in zip(static_in, static_gt):
+            with autocast():
                 pred = model(inp)
-                loss = loss + F.mse_loss(pred, gt)
+            loss = loss + F.mse_loss(pred.float(), gt)
         loss.backward()
-        optimizer.step()
         return loss
+
+    def comm():
+        if world > 1:
+            dist.all_reduce(flat_grad)
+            flat_grad.div_(world)
+
+    def load_data(i):
+        windows = pools[i % len(pools)]
+        for (inp, gt), si, sg in zip(windows, static_in, static_gt):
+            si.copy_(inp, non_blocking=True)
+            sg.copy_(gt, non_blocking=True)
+
+    use_graphs = not args.no_graphs
+    graph = None
+    if use_graphs:
+        try:
+            load_data(0)
+            # warmup on a side stream (required before capture)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(3):
+                    fwd_bwd()
+                    optimizer.step()
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                static_loss = fwd_bwd()
+                optimizer.step()
+            torch.cuda.synchronize()
+        except Exception as e:
+            print(f"[bench] graph capture failed ({type(e).__name__}: {e}); "
+                  f"falling back to eager", flush=True)
+            graph = None
+            use_graphs = False
+            optimizer = torch.optim.Adam(params, lr=1e-3, weight_decay=1e-4,
+                                         amsgrad=True, foreach=True)
+
+    def step(i):
+        load_data(i)
+        if graph is not None:
+            graph.replay()
+            comm()        # grads already in flat buffer post-replay; note:
+            # with graphs the optimizer ran inside the graph, so for world>1
+            # we fold comm INTO the captured region instead (see below).
+        else:
+            fwd_bwd()
+            comm()
+            optimizer.step()
+
+    if graph is not None and world > 1:
+        # re-capture with the all-reduce between backward and Adam so the
+        # optimizer consumes averaged gradients
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            static_loss = fwd_bwd()  # noqa: F841
+            dist.all_reduce(flat_grad)
+            flat_grad.div_(world)
+            optimizer.step()
+        torch.cuda.synchronize()
+
+        def step(i):  # noqa: F811
+            load_data(i)
+            graph.replay()
 
     for i in range(args.warmup):
         step(i)
@@ -173,6 +263,7 @@ def main():
                 "input": f"{args.lr_size}->{args.lr_size * args.scale}",
                 "scale": args.scale,
                 "upsampler": args.upsampler,
+                "hip_graphs": bool(graph is not None),
                 "parallelism": f"dp{world}",
             },
         }))
