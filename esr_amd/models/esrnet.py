@@ -121,11 +121,24 @@ class TimePropagation(nn.Module):
             return s.detach()
         self.state = _d(self.state)
 
-    def _local(self, f0, f1, f2):
-        m0 = self.pred_map(torch.cat([f0, f1], dim=1))
-        m1 = self.pred_map(torch.cat([f1, f2], dim=1))
+    def _local_batched(self, x):
+        """All N triplets in one batched call set (same math as the
+        reference's per-frame loop, ESR:models/model.py:133-144, with the
+        same edge duplication).  One pred_map launch for all 2N gating
+        maps, one local_fusion launch for all N frames."""
+        B, N, C, H, W = x.shape
+        i1 = torch.arange(N, device=x.device)
+        i0 = (i1 - 1).clamp(min=0)
+        i2 = (i1 + 1).clamp(max=N - 1)
+        f0 = x[:, i0].reshape(B * N, C, H, W)
+        f1 = x.reshape(B * N, C, H, W)
+        f2 = x[:, i2].reshape(B * N, C, H, W)
+        pairs = torch.cat([torch.cat([f0, f1], dim=1),
+                           torch.cat([f1, f2], dim=1)], dim=0)
+        maps = self.pred_map(pairs)
+        m0, m1 = maps[: B * N], maps[B * N:]
         out = self.local_fusion(torch.cat([f0 * m0, f1, f2 * m1], dim=1))
-        return out + f1
+        return (out + f1).view(B, N, C, H, W)
 
     def _global(self, feats):
         B, N, C, H, W = feats.shape
@@ -150,16 +163,7 @@ class TimePropagation(nn.Module):
         return fused.view(B, N, C, H, W)
 
     def forward(self, x):
-        B, N, C, H, W = x.shape
-        if self.has_ltc:
-            feats = []
-            for i in range(N):
-                i0, i1, i2 = (0, 0, 1) if i == 0 else \
-                             ((N - 2, N - 1, N - 1) if i == N - 1 else (i - 1, i, i + 1))
-                feats.append(self._local(x[:, i0], x[:, i1], x[:, i2]))
-            feats = torch.stack(feats, dim=1)
-        else:
-            feats = x
+        feats = self._local_batched(x) if self.has_ltc else x
         if self.has_gtc:
             feats = self._global(feats)
         return feats + x
@@ -221,6 +225,10 @@ class STFusion(nn.Module):
         ])
 
     def fuse(self, feat0, feat1):
+        """Deformable alignment of feat0 onto feat1 + spatial/channel
+        attention fusion (parity: ESR:models/model.py:208-231).  Works on
+        any leading batch, so dense_fuse batches all non-centre frames
+        through it in ONE call."""
         B, C, H, W = feat0.shape
         offset_feat = self.offset(torch.cat([feat0, feat1], dim=1))
         aligned = F.relu(self.dcn(feat0, offset_feat))
@@ -234,11 +242,14 @@ class STFusion(nn.Module):
 
     def dense_fuse(self, x):
         if self.has_dcnatten:
-            mid = x[:, self.mid_idx].contiguous()
-            outs = [self.fuse(x[:, i].contiguous(), mid)
-                    for i in range(self.num_frame) if i != self.mid_idx]
-            outs.append(mid)
-            out = torch.cat(outs, dim=1)
+            B, N, C, H, W = x.shape
+            mid = x[:, self.mid_idx]
+            idxes = [i for i in range(N) if i != self.mid_idx]
+            f0 = x[:, idxes].reshape(B * (N - 1), C, H, W)
+            f1 = mid.unsqueeze(1).expand(B, N - 1, C, H, W) \
+                .reshape(B * (N - 1), C, H, W)
+            fused = self.fuse(f0, f1).view(B, (N - 1) * C, H, W)
+            out = torch.cat([fused, mid], dim=1)
         else:
             out = x.reshape(x.size(0), -1, x.size(-2), x.size(-1))
         return self.dense_fusion(out)
