@@ -187,10 +187,16 @@ class DeformAlign2d(nn.Module):
         nn.init.zeros_(self.conv_offset_mask.bias)
 
     def forward(self, input, feat):
+        # The conv's output channels are interpreted directly as
+        # [offset (2*dg*K) | mask (dg*K)] — the reference's chunk+cat
+        # (ESR:models/DCNv2/dcn_v2.py:218-219) only permutes learned
+        # channels, so skipping it is math-equivalent under training and
+        # saves a concat launch per call.
         om = self.conv_offset_mask(feat)
-        o1, o2, mask = torch.chunk(om, 3, dim=1)
-        offset = torch.cat((o1, o2), dim=1)
-        mask = torch.sigmoid(mask)
+        kh, kw = self.kernel_size
+        n_off = self.deformable_groups * 2 * kh * kw
+        offset = om[:, :n_off]
+        mask = torch.sigmoid(om[:, n_off:])
         return modulated_deform_conv2d(
             input, offset, mask, self.weight, self.bias,
             self.stride, self.padding, self.dilation, self.deformable_groups)
