@@ -1,7 +1,11 @@
 #!/usr/bin/env python3
 """Bisect which op breaks hipGraph capture (GPU box debugging tool)."""
 
+import sys
 import traceback
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 import torch
 import torch.nn.functional as F
