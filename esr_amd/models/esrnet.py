@@ -142,8 +142,9 @@ class TimePropagation(nn.Module):
 
     def _global(self, feats):
         B, N, C, H, W = feats.shape
-        rev_idx = list(reversed(range(N)))
-        rev = feats[:, rev_idx]
+        # time reversal via torch.flip (hipGraph-capture-safe; python-list
+        # advanced indexing does a blocking H2D index copy)
+        rev = torch.flip(feats, dims=[1])
         state = None if self.gtc_frozen else self.state
         outs = []
         for i in range(N):
@@ -157,7 +158,7 @@ class TimePropagation(nn.Module):
 
         both = torch.stack(outs, dim=1)                               # [2B,N,...]
         x = both[:B]
-        r = both[B:][:, rev_idx]
+        r = torch.flip(both[B:], dims=[1])
         fused = torch.cat([x, r], dim=2).view(B * N, 2 * C, H, W)
         fused = self.global_fusion(fused)
         return fused.view(B, N, C, H, W)
@@ -244,8 +245,9 @@ class STFusion(nn.Module):
         if self.has_dcnatten:
             B, N, C, H, W = x.shape
             mid = x[:, self.mid_idx]
-            idxes = [i for i in range(N) if i != self.mid_idx]
-            f0 = x[:, idxes].reshape(B * (N - 1), C, H, W)
+            # capture-safe non-centre selection (slices, no index tensors)
+            f0 = torch.cat([x[:, :self.mid_idx], x[:, self.mid_idx + 1:]],
+                           dim=1).reshape(B * (N - 1), C, H, W)
             f1 = mid.unsqueeze(1).expand(B, N - 1, C, H, W) \
                 .reshape(B * (N - 1), C, H, W)
             fused = self.fuse(f0, f1).view(B, (N - 1) * C, H, W)
