@@ -126,9 +126,10 @@ def main():
                              lr_res, hr_res, device, ext, seed=100 + rank)
     amp_dtype = torch.bfloat16 if args.dtype == "bf16" else None
 
-    # static input buffers (graph replay reads from fixed addresses)
-    static_in = [torch.empty_like(inp) for inp, _ in pools[0]]
-    static_gt = [torch.empty_like(gt) for _, gt in pools[0]]
+    # graph replay reads from fixed addresses: one graph per pre-generated
+    # data set (no per-step copies), all sharing one memory pool
+    static_in = [inp for inp, _ in pools[0]]
+    static_gt = [gt for _, gt in pools[0]]
 
     import contextlib
 
@@ -155,66 +156,54 @@ def main():
             dist.all_reduce(flat_grad)
             flat_grad.div_(world)
 
-    def load_data(i):
+    def set_data(i):
         windows = pools[i % len(pools)]
-        for (inp, gt), si, sg in zip(windows, static_in, static_gt):
-            si.copy_(inp, non_blocking=True)
-            sg.copy_(gt, non_blocking=True)
+        for w, (inp, gt) in enumerate(windows):
+            static_in[w] = inp
+            static_gt[w] = gt
 
     use_graphs = not args.no_graphs
-    graph = None
+    graphs = None
     if use_graphs:
         try:
-            load_data(0)
             # warmup on a side stream (required before capture)
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(s):
                 for _ in range(3):
                     fwd_bwd()
+                    comm()
                     optimizer.step()
             torch.cuda.current_stream().wait_stream(s)
             torch.cuda.synchronize()
-            graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                static_loss = fwd_bwd()
-                optimizer.step()
+            graphs = []
+            pool_handle = None
+            for i in range(len(pools)):
+                set_data(i)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g, pool=pool_handle):
+                    fwd_bwd()
+                    comm()            # RCCL all-reduce is graph-capturable
+                    optimizer.step()
+                pool_handle = g.pool()
+                graphs.append(g)
             torch.cuda.synchronize()
         except Exception as e:
             print(f"[bench] graph capture failed ({type(e).__name__}: {e}); "
                   f"falling back to eager", flush=True)
-            graph = None
+            graphs = None
             use_graphs = False
             optimizer = torch.optim.Adam(params, lr=1e-3, weight_decay=1e-4,
                                          amsgrad=True, foreach=True)
 
     def step(i):
-        load_data(i)
-        if graph is not None:
-            graph.replay()
-            comm()        # grads already in flat buffer post-replay; note:
-            # with graphs the optimizer ran inside the graph, so for world>1
-            # we fold comm INTO the captured region instead (see below).
+        if graphs is not None:
+            graphs[i % len(graphs)].replay()
         else:
+            set_data(i)
             fwd_bwd()
             comm()
             optimizer.step()
-
-    if graph is not None and world > 1:
-        # re-capture with the all-reduce between backward and Adam so the
-        # optimizer consumes averaged gradients
-        torch.cuda.synchronize()
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
-            static_loss = fwd_bwd()  # noqa: F841
-            dist.all_reduce(flat_grad)
-            flat_grad.div_(world)
-            optimizer.step()
-        torch.cuda.synchronize()
-
-        def step(i):  # noqa: F811
-            load_data(i)
-            graph.replay()
 
     for i in range(args.warmup):
         step(i)
@@ -263,7 +252,7 @@ def main():
                 "input": f"{args.lr_size}->{args.lr_size * args.scale}",
                 "scale": args.scale,
                 "upsampler": args.upsampler,
-                "hip_graphs": bool(graph is not None),
+                "hip_graphs": bool(graphs is not None),
                 "parallelism": f"dp{world}",
             },
         }))
