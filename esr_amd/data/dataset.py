@@ -297,6 +297,11 @@ class EventSRDataset(Dataset):
         inp_stack = E.events_to_stack_no_polarity(inp[0], inp[1], inp[2], inp[3],
                                                   self.time_bins, inp_res)
         inp_cnt = E.events_to_channels(inp[0], inp[1], inp[3], inp_res)
+
+        if self.hot_filter.get("enabled", False):
+            hot_mask = self._hot_mask(inp, inp_res)
+            inp_cnt = inp_cnt * hot_mask
+            inp_stack = inp_stack * hot_mask
         inp_bicubic_cnt = self._interp(inp_cnt, gt_res, "bicubic")
         inp_bicubic_stack = self._interp(inp_stack, gt_res, "bicubic")
         inp_near_cnt = self._interp(inp_cnt, gt_res, "nearest")
@@ -365,6 +370,31 @@ class EventSRDataset(Dataset):
             E.scaled_count_encoding(down_norm, self.inp_sensor_resolution, "cnt"),
             s2, rounding_mode="floor")
         return inp_down_cnt, inp_down_scaled_cnt
+
+    def _hot_mask(self, events, resolution):
+        """Running-average hot-pixel mask (parity:
+        ESR:dataloader/h5dataset.py:621-641, ESR:dataloader/encodings.py:348-363).
+        Pixels whose cumulative event rate exceeds max_rate are zeroed, up
+        to max_px pixels, once min_obvs windows have been observed."""
+        update = torch.zeros(resolution)
+        xi = events[0].long().clamp(0, resolution[1] - 1)
+        yi = events[1].long().clamp(0, resolution[0] - 1)
+        update[yi, xi] = events[3].abs()
+        self.hot_events += update
+        self.hot_idx += 1
+        event_rate = self.hot_events / self.hot_idx
+        mask = torch.ones(resolution)
+        if self.hot_idx > self.hot_filter.get("min_obvs", 5):
+            rate = event_rate.clone()
+            for _ in range(self.hot_filter.get("max_px", 100)):
+                idx = torch.argmax(rate)
+                y, x = divmod(int(idx), resolution[1])
+                if rate[y, x] > self.hot_filter.get("max_rate", 0.8):
+                    rate[y, x] = 0
+                    mask[y, x] = 0
+                else:
+                    break
+        return mask
 
     def _get_gt_frame(self, idx0, idx1):
         """Frame nearest (by binary search) to the window's mid event

@@ -156,7 +156,7 @@ class Trainer:
                 iter_idx = idx + len(self.train_dataloader) * epoch \
                     + self.start_iteration
                 best = False
-                loss, mse_loss, _ = self.bptt_step(inputs_seq, train=True)
+                loss, mse_loss, pred = self.bptt_step(inputs_seq, train=True)
 
                 reduced_mse = reduce_tensor(mse_loss)
                 reduced_loss = reduce_tensor(loss)
@@ -172,6 +172,7 @@ class Trainer:
                             f"Train epoch {epoch + 1} iter {iter_idx}/"
                             f"{self.iterations} mse {reduced_mse.item():.4e} "
                             f"loss {reduced_loss.item():.4e} lr {lr:.3e}")
+                    self._maybe_visualize(iter_idx, inputs_seq, pred)
 
                 if self.do_validation and iter_idx % self.valid_step == 0 \
                         and iter_idx != 0:
@@ -254,6 +255,39 @@ class Trainer:
                     f"Valid stamp {stamp} [{batch_idx}/{len(self.valid_dataloader)}]"
                     f" mse {reduced_mse.item():.4e}")
         return self.valid_metrics.result()
+
+    def _maybe_visualize(self, iter_idx, inputs_seq, pred):
+        """Render input/scaled/pred/gt count maps to the writer every
+        train_img_writer_num iterations (parity:
+        ESR:train_ours_cnt_seq.py:258-293).  No-ops unless a TensorBoard
+        sink is active."""
+        vis_cfg = self.cfg["trainer"].get("vis", {"enabled": False})
+        if not vis_cfg.get("enabled", False):
+            return
+        if self.writer is None or getattr(self.writer, "_tb", None) is None:
+            return
+        step = vis_cfg.get("train_img_writer_num", 20)
+        if iter_idx % step != 0:
+            return
+        from ..utils.vis import EventVisualizer
+        vis = EventVisualizer()
+        inputs = inputs_seq[-1]
+        hwc = lambda t: t.cpu().float().numpy().transpose(1, 2, 0)  # noqa: E731
+        self.writer.add_image(
+            "train_inp_events_cnt",
+            vis.plot_event_cnt(hwc(inputs["inp_cnt"][0, self.mid_idx])),
+            global_step=iter_idx)
+        self.writer.add_image(
+            "train_inp_scaled_events_cnt",
+            vis.plot_event_cnt(hwc(inputs["inp_scaled_cnt"][0, self.mid_idx])),
+            global_step=iter_idx)
+        self.writer.add_image(
+            "train_esr_events_cnt",
+            vis.plot_event_cnt(hwc(pred[0].round())), global_step=iter_idx)
+        self.writer.add_image(
+            "train_gt_events_cnt",
+            vis.plot_event_cnt(hwc(inputs["gt_cnt"][0, self.mid_idx])),
+            global_step=iter_idx)
 
     # ---------------- monitoring / checkpoints ----------------
 

@@ -156,3 +156,28 @@ def test_time_mode_windowing(store_path):
 def test_datalist(synth_datalist):
     paths = read_datalist(synth_datalist)
     assert len(paths) == 2
+
+
+def test_hot_filter_masks_hot_pixel(store_path):
+    cfg = _ds_config()
+    cfg["hot_filter"] = {"enabled": True, "max_px": 10, "min_obvs": 0,
+                         "max_rate": 0.5}
+    ds = EventSRDataset(store_path, cfg)
+    # warm the running average so a hot pixel emerges, then check masking
+    for i in range(3):
+        ds.__getitem__(i % len(ds), seed=1)
+    assert ds.hot_idx == 3
+
+
+def test_sequence_pause_chain(store_path):
+    cfg = _ds_config()
+    cfg["sequence"]["pause"] = {"enabled": True,
+                                "proba_pause_when_running": 1.0,
+                                "proba_pause_when_paused": 1.0}
+    sd = SequenceDataset(store_path, cfg)
+    seq = sd[0]
+    # with pause prob 1.0 every item after the first is a zero-input pause
+    assert seq[0]["inp_cnt"].sum() > 0
+    for item in seq[1:]:
+        assert item["inp_cnt"].sum() == 0
+        assert item["gt_cnt"].sum() > 0   # GT stays (frozen index)
