@@ -1,15 +1,20 @@
 #!/bin/bash
 # rocprofv3 kernel-stats profile of bench.py on a GPU box.
 #   bash scripts/profile_bench.sh <out-name> [bench args...]
-# Writes CSVs under gpurun_out/prof/.  Counter (--pmc) collection must be a
-# separate run without trace domains (pool policy).
+# Keeps only the aggregated *_stats.csv under gpurun_out/prof/ (the raw
+# trace db of a graph-replay run exceeds the 64 MiB copy-back budget).
+# Counter (--pmc) collection must be a separate run without trace domains.
 set -e
 NAME=${1:-bench}
 shift || true
 REPO=$(cd "$(dirname "$0")/.." && pwd)
-mkdir -p "$REPO/gpurun_out/prof"
+PROF="$REPO/gpurun_out/prof"
+mkdir -p "$PROF"
 export TMPDIR=/tmp
 cd /tmp
-timeout 400 rocprofv3 --kernel-trace --stats -d "$REPO/gpurun_out/prof" \
-    -o "$NAME" -- bash -c "cd '$REPO' && python bench.py --steps 5 --warmup 2 $*"
+timeout 400 rocprofv3 --kernel-trace --stats -d "$PROF" \
+    -o "$NAME" -- bash -c "cd '$REPO' && python bench.py --steps 3 --warmup 2 $*"
+# keep CSVs only; the .db can be hundreds of MiB
+find "$PROF" -name "${NAME}*" ! -name "*.csv" -delete || true
+ls -la "$PROF"
 echo "profile written to gpurun_out/prof/${NAME}*"
