@@ -162,43 +162,66 @@ def main():
             static_in[w] = inp
             static_gt[w] = gt
 
-    use_graphs = not args.no_graphs
+    def _warmup_side_stream(body):
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                body()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+
+    def _capture(body):
+        graphs = []
+        pool_handle = None
+        for i in range(len(pools)):
+            set_data(i)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g, pool=pool_handle):
+                body()
+            pool_handle = g.pool()
+            graphs.append(g)
+        torch.cuda.synchronize()
+        return graphs
+
+    # Tiered capture: (a) whole step incl. RCCL all-reduce + Adam in the
+    # graph; (b) fwd+bwd in the graph, comm + Adam eager; (c) fully eager.
     graphs = None
-    if use_graphs:
+    graph_mode = "eager"
+    if not args.no_graphs:
+        def full_step():
+            fwd_bwd()
+            comm()
+            optimizer.step()
         try:
-            # warmup on a side stream (required before capture)
-            s = torch.cuda.Stream()
-            s.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(s):
-                for _ in range(3):
-                    fwd_bwd()
-                    comm()
-                    optimizer.step()
-            torch.cuda.current_stream().wait_stream(s)
-            torch.cuda.synchronize()
-            graphs = []
-            pool_handle = None
-            for i in range(len(pools)):
-                set_data(i)
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g, pool=pool_handle):
-                    fwd_bwd()
-                    comm()            # RCCL all-reduce is graph-capturable
-                    optimizer.step()
-                pool_handle = g.pool()
-                graphs.append(g)
-            torch.cuda.synchronize()
+            _warmup_side_stream(full_step)
+            graphs = _capture(full_step)
+            graph_mode = "full"
         except Exception as e:
-            print(f"[bench] graph capture failed ({type(e).__name__}: {e}); "
-                  f"falling back to eager", flush=True)
-            graphs = None
-            use_graphs = False
+            print(f"[bench] full-step capture failed ({type(e).__name__}); "
+                  f"trying fwd+bwd-only capture", flush=True)
+            torch.cuda.synchronize()
+            try:
+                _warmup_side_stream(fwd_bwd)
+                graphs = _capture(fwd_bwd)
+                graph_mode = "fwd_bwd"
+            except Exception as e2:
+                print(f"[bench] capture failed ({type(e2).__name__}: {e2}); "
+                      f"eager fallback", flush=True)
+                graphs = None
+        if graph_mode != "full":
+            # capturable Adam steps on-device tensors; for eager modes use a
+            # plain (faster host-side) Adam
             optimizer = torch.optim.Adam(params, lr=1e-3, weight_decay=1e-4,
                                          amsgrad=True, foreach=True)
 
     def step(i):
-        if graphs is not None:
+        if graph_mode == "full":
             graphs[i % len(graphs)].replay()
+        elif graph_mode == "fwd_bwd":
+            graphs[i % len(graphs)].replay()
+            comm()
+            optimizer.step()
         else:
             set_data(i)
             fwd_bwd()
@@ -252,7 +275,7 @@ def main():
                 "input": f"{args.lr_size}->{args.lr_size * args.scale}",
                 "scale": args.scale,
                 "upsampler": args.upsampler,
-                "hip_graphs": bool(graphs is not None),
+                "hip_graphs": graph_mode,
                 "parallelism": f"dp{world}",
             },
         }))
