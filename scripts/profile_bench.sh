@@ -12,7 +12,7 @@ PROF="$REPO/gpurun_out/prof"
 mkdir -p "$PROF"
 export TMPDIR=/tmp
 cd /tmp
-timeout 400 rocprofv3 --kernel-trace --stats -d "$PROF" \
+timeout 400 rocprofv3 --kernel-trace --stats --output-format csv -d "$PROF" \
     -o "$NAME" -- bash -c "cd '$REPO' && python bench.py --steps 3 --warmup 2 $*"
 # keep CSVs only; the .db can be hundreds of MiB
 find "$PROF" -name "${NAME}*" ! -name "*.csv" -delete || true
