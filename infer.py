@@ -39,7 +39,9 @@ def default_dataloader_config(args):
 
 def main():
     p = argparse.ArgumentParser()
-    p.add_argument("--model_path", type=str, required=True)
+    p.add_argument("--model_path", type=str, default=None)
+    p.add_argument("--model_list", type=str, default=None,
+                   help="txt of checkpoint paths; evaluates each")
     p.add_argument("--data_path", type=str, default=None)
     p.add_argument("--data_list", type=str, default=None)
     p.add_argument("--device", type=str, default="cuda:0")
@@ -63,9 +65,14 @@ def main():
 
     device = torch.device(args.device if torch.cuda.is_available() else "cpu")
     dl_cfg = default_dataloader_config(args)
-    model, _ = load_model_from_checkpoint(args.model_path, device=device,
-                                          seqn=args.seqn)
     metrics = build_metrics(device, lpips_weights=args.lpips_weights)
+
+    if args.model_list:
+        model_paths = read_datalist(args.model_list)
+    elif args.model_path:
+        model_paths = [args.model_path]
+    else:
+        raise SystemExit("provide --model_path or --model_list")
 
     if args.data_list:
         paths = read_datalist(args.data_list)
@@ -75,23 +82,31 @@ def main():
         raise SystemExit("provide --data_path or --data_list")
 
     out_root = Path(args.output_path)
-    all_results = {}
-    for path in paths:
-        name = Path(path).stem
-        res = infer_sequence(dl_cfg, path, model, device,
-                             output_path=out_root / name, metrics=metrics,
-                             save_images=not args.no_images)
-        all_results[name] = res
-        print(f"{name}: {res}")
+    out_root.mkdir(parents=True, exist_ok=True)
+    for model_path in model_paths:
+        model, _ = load_model_from_checkpoint(model_path, device=device,
+                                              seqn=args.seqn)
+        mdir = out_root if len(model_paths) == 1 \
+            else out_root / Path(model_path).stem
+        all_results = {}
+        for path in paths:
+            name = Path(path).stem
+            res = infer_sequence(dl_cfg, path, model, device,
+                                 output_path=mdir / name, metrics=metrics,
+                                 save_images=not args.no_images)
+            all_results[name] = res
+            print(f"{Path(model_path).stem} / {name}: {res}")
 
-    mean = {}
-    if all_results:
-        keys = next(iter(all_results.values())).keys()
-        for k in keys:
-            mean[k] = sum(r[k] for r in all_results.values()) / len(all_results)
-    with open(out_root / "mean_results.yml", "w") as f:
-        yaml.safe_dump({"per_file": all_results, "mean": mean}, f)
-    print("mean:", mean)
+        mean = {}
+        if all_results:
+            keys = next(iter(all_results.values())).keys()
+            for k in keys:
+                mean[k] = sum(r[k] for r in all_results.values()) \
+                    / len(all_results)
+        mdir.mkdir(parents=True, exist_ok=True)
+        with open(mdir / "mean_results.yml", "w") as f:
+            yaml.safe_dump({"per_file": all_results, "mean": mean}, f)
+        print("mean:", mean)
 
 
 if __name__ == "__main__":
