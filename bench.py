@@ -85,6 +85,12 @@ def main():
     p.add_argument("--upsampler", type=str, default="pixelshuffle")
     p.add_argument("--no-graphs", action="store_true",
                    help="disable hipGraph capture (eager mode)")
+    p.add_argument("--channels-last", action="store_true",
+                   help="NHWC weights/activations (skips MIOpen transposes)")
+    p.add_argument("--cast-mode", choices=["autocast", "pure"],
+                   default="autocast",
+                   help="bf16 via autocast, or pure bf16 weights with an "
+                        "fp32 master-weight Adam (no per-op casts)")
     args = p.parse_args()
 
     assert torch.cuda.is_available(), "bench.py requires an MI355X GPU"
@@ -105,19 +111,40 @@ def main():
     model = build_model("ESRNet", inch=2, basech=args.basech,
                         num_frame=args.seqn,
                         upsampler=args.upsampler).to(device)
+    if args.channels_last:
+        model = model.to(memory_format=torch.channels_last)
 
-    # one flat fp32 gradient buffer -> a single RCCL all-reduce per step
+    pure_bf16 = args.cast_mode == "pure" and args.dtype == "bf16"
+    if pure_bf16:
+        model = model.to(torch.bfloat16)
+
+    # one flat gradient buffer -> a single RCCL all-reduce per step
     params = [prm for prm in model.parameters() if prm.requires_grad]
     total = sum(prm.numel() for prm in params)
-    flat_grad = torch.zeros(total, device=device)
+    flat_grad = torch.zeros(total, device=device,
+                            dtype=torch.bfloat16 if pure_bf16 else torch.float32)
     off = 0
     for prm in params:
         prm.grad = flat_grad[off:off + prm.numel()].view_as(prm)
         off += prm.numel()
 
-    optimizer = torch.optim.Adam(params, lr=1e-3, weight_decay=1e-4,
-                                 amsgrad=True, foreach=True,
-                                 capturable=not args.no_graphs)
+    if pure_bf16:
+        # fp32 master weights; Adam runs on masters, bf16 working copy is
+        # refreshed once per step (one foreach cast instead of per-op casts)
+        masters = [prm.detach().float().clone() for prm in params]
+        master_flat_grad = torch.zeros(total, device=device)
+        off = 0
+        for mprm in masters:
+            mprm.grad = master_flat_grad[off:off + mprm.numel()].view_as(mprm)
+            off += mprm.numel()
+        optimizer = torch.optim.Adam(masters, lr=1e-3, weight_decay=1e-4,
+                                     amsgrad=True, foreach=True,
+                                     capturable=not args.no_graphs)
+    else:
+        masters = None
+        optimizer = torch.optim.Adam(params, lr=1e-3, weight_decay=1e-4,
+                                     amsgrad=True, foreach=True,
+                                     capturable=not args.no_graphs)
 
     lr_res = (args.lr_size, args.lr_size)
     hr_res = (args.lr_size * args.scale, args.lr_size * args.scale)
@@ -134,7 +161,7 @@ def main():
     import contextlib
 
     def autocast():
-        if amp_dtype is None:
+        if amp_dtype is None or pure_bf16:
             return contextlib.nullcontext()
         # cache_enabled=False: the autocast weight-cast cache is not
         # graph-capture-safe (casts must be recorded into the graph)
@@ -145,8 +172,9 @@ def main():
         model.reset_states()
         loss = 0
         for inp, gt in zip(static_in, static_gt):
+            x = inp.to(torch.bfloat16) if pure_bf16 else inp
             with autocast():
-                pred = model(inp)
+                pred = model(x)
             loss = loss + F.mse_loss(pred.float(), gt)
         loss.backward()
         return loss
@@ -155,6 +183,14 @@ def main():
         if world > 1:
             dist.all_reduce(flat_grad)
             flat_grad.div_(world)
+
+    def opt_step():
+        if pure_bf16:
+            master_flat_grad.copy_(flat_grad)       # one bf16->fp32 cast
+            optimizer.step()
+            torch._foreach_copy_(params, masters)   # refresh bf16 weights
+        else:
+            optimizer.step()
 
     def set_data(i):
         windows = pools[i % len(pools)]
@@ -192,7 +228,7 @@ def main():
         def full_step():
             fwd_bwd()
             comm()
-            optimizer.step()
+            opt_step()
         try:
             _warmup_side_stream(full_step)
             graphs = _capture(full_step)
@@ -212,7 +248,8 @@ def main():
         if graph_mode != "full":
             # capturable Adam steps on-device tensors; for eager modes use a
             # plain (faster host-side) Adam
-            optimizer = torch.optim.Adam(params, lr=1e-3, weight_decay=1e-4,
+            optimizer = torch.optim.Adam(masters if pure_bf16 else params,
+                                         lr=1e-3, weight_decay=1e-4,
                                          amsgrad=True, foreach=True)
 
     def step(i):
@@ -221,12 +258,12 @@ def main():
         elif graph_mode == "fwd_bwd":
             graphs[i % len(graphs)].replay()
             comm()
-            optimizer.step()
+            opt_step()
         else:
             set_data(i)
             fwd_bwd()
             comm()
-            optimizer.step()
+            opt_step()
 
     for i in range(args.warmup):
         step(i)
