@@ -188,7 +188,8 @@ def main():
         if pure_bf16:
             master_flat_grad.copy_(flat_grad)       # one bf16->fp32 cast
             optimizer.step()
-            torch._foreach_copy_(params, masters)   # refresh bf16 weights
+            with torch.no_grad():
+                torch._foreach_copy_(params, masters)  # refresh bf16 weights
         else:
             optimizer.step()
 

@@ -138,7 +138,7 @@ class TimePropagation(nn.Module):
         maps = self.pred_map(pairs)
         m0, m1 = maps[: B * N], maps[B * N:]
         out = self.local_fusion(torch.cat([f0 * m0, f1, f2 * m1], dim=1))
-        return (out + f1).view(B, N, C, H, W)
+        return (out + f1).reshape(B, N, C, H, W)
 
     def _global(self, feats):
         B, N, C, H, W = feats.shape
@@ -159,9 +159,9 @@ class TimePropagation(nn.Module):
         both = torch.stack(outs, dim=1)                               # [2B,N,...]
         x = both[:B]
         r = torch.flip(both[B:], dims=[1])
-        fused = torch.cat([x, r], dim=2).view(B * N, 2 * C, H, W)
+        fused = torch.cat([x, r], dim=2).reshape(B * N, 2 * C, H, W)
         fused = self.global_fusion(fused)
-        return fused.view(B, N, C, H, W)
+        return fused.reshape(B, N, C, H, W)
 
     def forward(self, x):
         feats = self._local_batched(x) if self.has_ltc else x
@@ -235,7 +235,7 @@ class STFusion(nn.Module):
         aligned = F.relu(self.dcn(feat0, offset_feat))
         feat = self.convblock(torch.cat([aligned, feat1], dim=1))
         spatial_k = self.kernel(feat)                                  # [B,2,H,W]
-        pooled = feat.view(B, C, H * W).transpose(1, 2).max(1, keepdim=True)[0]
+        pooled = feat.reshape(B, C, H * W).transpose(1, 2).max(1, keepdim=True)[0]
         channel_k = self.fc(pooled).transpose(1, 2).unsqueeze(-1)      # [B,2C,1,1]
         y0 = aligned * spatial_k[:, :1] * channel_k[:, :C]
         y1 = feat1 * spatial_k[:, 1:2] * channel_k[:, C:]
@@ -250,7 +250,7 @@ class STFusion(nn.Module):
                            dim=1).reshape(B * (N - 1), C, H, W)
             f1 = mid.unsqueeze(1).expand(B, N - 1, C, H, W) \
                 .reshape(B * (N - 1), C, H, W)
-            fused = self.fuse(f0, f1).view(B, (N - 1) * C, H, W)
+            fused = self.fuse(f0, f1).reshape(B, (N - 1) * C, H, W)
             out = torch.cat([fused, mid], dim=1)
         else:
             out = x.reshape(x.size(0), -1, x.size(-2), x.size(-1))
@@ -261,7 +261,7 @@ class STFusion(nn.Module):
             B, N, C, H, W = feats.shape
             flat = feats.reshape(B * N, C, H, W)
             flat = flat * self.attens[idx](flat)
-            x = x + flat.view(B, N, C, H, W).mean(1)
+            x = x + flat.reshape(B, N, C, H, W).mean(1)
         return self.recons[idx](x)
 
     def forward(self, x, feats_list):
@@ -269,7 +269,7 @@ class STFusion(nn.Module):
         assert N == self.num_frame
         out = self.dense_fuse(x)
         for idx, feats in enumerate(feats_list):
-            feats = feats.view(B, N, -1, feats.size(-2), feats.size(-1))
+            feats = feats.reshape(B, N, -1, feats.size(-2), feats.size(-1))
             out = self.scale_aggre(out, feats, idx)
         return out
 
@@ -321,7 +321,7 @@ class ESRNet(nn.Module):
         x = self.head(x)
         feats_list = self.feat_extract(x)
         deep = feats_list[0]
-        deep = deep.view(B, N, -1, deep.size(-2), deep.size(-1))
+        deep = deep.reshape(B, N, -1, deep.size(-2), deep.size(-1))
         deep = self.time_propagate(deep)
         out = self.spacetime_fuse(deep, feats_list)
         out = self.tail(out)
