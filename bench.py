@@ -73,7 +73,7 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch", type=int, default=32, help="per-GPU batch")
+    p.add_argument("--batch", type=int, default=64, help="per-GPU batch")
     p.add_argument("--seql", type=int, default=8)
     p.add_argument("--seqn", type=int, default=3)
     p.add_argument("--window", type=int, default=2048)
