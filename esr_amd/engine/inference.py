@@ -20,8 +20,8 @@ import torch.nn.functional as F
 import yaml
 
 from ..data import InferenceSequenceDataLoader
-from ..loss import PerceptualLoss, l1 as l1_fn, mse as mse_fn, psnr as psnr_fn, \
-    ssim as ssim_fn
+from ..loss import PerceptualLoss, l1 as l1_fn, mse as mse_fn, \
+    psnr as psnr_fn, rmse as rmse_fn, ssim as ssim_fn
 from ..utils import MetricTracker
 from ..utils.vis import EventVisualizer
 
@@ -51,9 +51,9 @@ def infer_sequence(dataloader_config, data_path, model, device,
     dataloader = InferenceSequenceDataLoader(data_path, dataloader_config)
     gt_res = dataloader.gt_sensor_resolution
 
-    keys = ["esr_l1", "esr_mse", "esr_lpips", "esr_ssim", "esr_psnr",
-            "bicubic_l1", "bicubic_mse", "bicubic_lpips", "bicubic_ssim",
-            "bicubic_psnr", "time", "params"]
+    keys = ["esr_l1", "esr_mse", "esr_rmse", "esr_lpips", "esr_ssim",
+            "esr_psnr", "bicubic_l1", "bicubic_mse", "bicubic_rmse",
+            "bicubic_lpips", "bicubic_ssim", "bicubic_psnr", "time", "params"]
     track = MetricTracker(keys)
 
     if output_path is not None:
@@ -106,6 +106,7 @@ def infer_sequence(dataloader_config, data_path, model, device,
         for name, pred in (("esr", esr_cnt), ("bicubic", bicubic_cnt)):
             track.update(f"{name}_l1", metrics["l1"](pred, gt_cnt).item())
             track.update(f"{name}_mse", metrics["mse"](pred, gt_cnt).item())
+            track.update(f"{name}_rmse", rmse_fn(pred, gt_cnt).item())
             track.update(f"{name}_ssim", metrics["ssim"](pred, gt_cnt))
             track.update(f"{name}_psnr", metrics["psnr"](pred, gt_cnt))
             track.update(f"{name}_lpips",

@@ -227,3 +227,37 @@ class TestModelGPU:
         y_gpu = mg(x.cuda())
         assert torch.allclose(y_cpu, y_gpu.cpu(), atol=5e-3), \
             (y_cpu - y_gpu.cpu()).abs().max().item()
+
+
+class TestBenchPath:
+    def test_bench_graph_step(self, ext):
+        """End-to-end bench step (graph capture + replay) as a subprocess."""
+        import json
+        import subprocess
+        import sys
+        out = subprocess.run(
+            [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+             "--batch", "2"],
+            capture_output=True, text=True, timeout=600,
+            cwd=str(__import__("pathlib").Path(__file__).parent.parent))
+        assert out.returncode == 0, out.stderr[-2000:]
+        line = [ln for ln in out.stdout.splitlines() if ln.startswith("{")][-1]
+        rec = json.loads(line)
+        assert rec["value"] > 0
+        assert rec["config"]["hip_graphs"] == "full"
+
+    def test_redistribute_on_gpu(self, ext):
+        from esr_amd.ops import redistribute_stack, events_to_stack_no_polarity
+        g = torch.Generator().manual_seed(0)
+        stack = torch.randint(-3, 8, (2, 4, 8, 8), generator=g).float().cuda()
+        cloud = redistribute_stack(stack, mode="linear")
+        assert cloud.is_cuda
+        # round trip on device
+        for b in range(2):
+            ev = cloud[b]
+            ev = ev[ev.abs().sum(1) > 0]
+            rebuilt = torch.zeros(4, 8, 8, device="cuda")
+            c = ((ev[:, 2] - 1e-6) * 4).long().clamp(0, 3)
+            idx = c * 64 + ev[:, 1].long() * 8 + ev[:, 0].long()
+            rebuilt.view(-1).scatter_add_(0, idx, ev[:, 3])
+            assert torch.allclose(rebuilt, stack[b])
