@@ -121,53 +121,55 @@ class TimePropagation(nn.Module):
             return s.detach()
         self.state = _d(self.state)
 
-    def _local_batched(self, x):
+    def _local_batched(self, x, N):
         """All N triplets in one batched call set (same math as the
         reference's per-frame loop, ESR:models/model.py:133-144, with the
-        same edge duplication).  One pred_map launch for all 2N gating
-        maps, one local_fusion launch for all N frames."""
-        B, N, C, H, W = x.shape
-        i1 = torch.arange(N, device=x.device)
-        i0 = (i1 - 1).clamp(min=0)
-        i2 = (i1 + 1).clamp(max=N - 1)
-        f0 = x[:, i0].reshape(B * N, C, H, W)
-        f1 = x.reshape(B * N, C, H, W)
-        f2 = x[:, i2].reshape(B * N, C, H, W)
-        pairs = torch.cat([torch.cat([f0, f1], dim=1),
-                           torch.cat([f1, f2], dim=1)], dim=0)
+        same edge duplication).  x is FRAME-major [N*B, C, H, W]: frame
+        shifts are contiguous slice concatenations (no index tensors, no 5D
+        round trips)."""
+        B = x.size(0) // N
+        f0 = torch.cat([x[:B], x[:-B]], dim=0)       # frame i-1 (dup frame 0)
+        f2 = torch.cat([x[B:], x[-B:]], dim=0)       # frame i+1 (dup frame N-1)
+        pairs = torch.cat([torch.cat([f0, x], dim=1),
+                           torch.cat([x, f2], dim=1)], dim=0)
         maps = self.pred_map(pairs)
-        m0, m1 = maps[: B * N], maps[B * N:]
-        out = self.local_fusion(torch.cat([f0 * m0, f1, f2 * m1], dim=1))
-        return (out + f1).reshape(B, N, C, H, W)
+        m0, m1 = maps[: N * B], maps[N * B:]
+        out = self.local_fusion(torch.cat([f0 * m0, x, f2 * m1], dim=1))
+        return out + x
 
-    def _global(self, feats):
-        B, N, C, H, W = feats.shape
-        # time reversal via torch.flip (hipGraph-capture-safe; python-list
-        # advanced indexing does a blocking H2D index copy)
-        rev = torch.flip(feats, dims=[1])
+    def _global(self, feats, N):
+        """Bi-directional shared-weight GRU scan over frame-major input;
+        fwd and bwd direction run in one batched cell call per step."""
+        B = feats.size(0) // N
         state = None if self.gtc_frozen else self.state
         outs = []
-        for i in range(N):
-            # fwd and bwd scan share weights -> run both in one batched call
-            step_in = torch.cat([feats[:, i], rev[:, i]], dim=0)     # [2B,...]
-            out, state = self.lstm(step_in, state)
+        for n in range(N):
+            fwd_in = feats[n * B:(n + 1) * B]
+            bwd_in = feats[(N - 1 - n) * B:(N - n) * B]
+            out, state = self.lstm(torch.cat([fwd_in, bwd_in], dim=0), state)
             if self.gtc_frozen:
                 state = None
-            outs.append(out)
+            outs.append(out)                          # [2B, C, H, W]
         self.state = None if self.gtc_frozen else state
 
-        both = torch.stack(outs, dim=1)                               # [2B,N,...]
-        x = both[:B]
-        r = torch.flip(both[B:], dims=[1])
-        fused = torch.cat([x, r], dim=2).reshape(B * N, 2 * C, H, W)
-        fused = self.global_fusion(fused)
-        return fused.reshape(B, N, C, H, W)
+        fused = torch.cat(
+            [torch.cat([outs[n][:B], outs[N - 1 - n][B:]], dim=1)
+             for n in range(N)], dim=0)               # [N*B, 2C, H, W]
+        return self.global_fusion(fused)
+
+    def forward_frames(self, x, N):
+        """Frame-major entry: x [N*B, C, H, W]."""
+        feats = self._local_batched(x, N) if self.has_ltc else x
+        if self.has_gtc:
+            feats = self._global(feats, N)
+        return feats + x
 
     def forward(self, x):
-        feats = self._local_batched(x) if self.has_ltc else x
-        if self.has_gtc:
-            feats = self._global(feats)
-        return feats + x
+        """5D-compat entry: x [B, N, C, H, W]."""
+        B, N, C, H, W = x.shape
+        xf = x.transpose(0, 1).reshape(N * B, C, H, W)
+        out = self.forward_frames(xf, N)
+        return out.reshape(N, B, C, H, W).transpose(0, 1)
 
 
 class STFusion(nn.Module):
@@ -241,37 +243,47 @@ class STFusion(nn.Module):
         y1 = feat1 * spatial_k[:, 1:2] * channel_k[:, C:]
         return self.dcn_fusion(torch.cat([y0, y1], dim=1))
 
-    def dense_fuse(self, x):
+    def dense_fuse_frames(self, x, N):
+        """x FRAME-major [N*B, C, H, W]; aligns every non-centre frame to
+        the centre in one batched DCN/attention pass."""
+        B = x.size(0) // N
+        m = self.mid_idx
+        mid = x[m * B:(m + 1) * B]
         if self.has_dcnatten:
-            B, N, C, H, W = x.shape
-            mid = x[:, self.mid_idx]
-            # capture-safe non-centre selection (slices, no index tensors)
-            f0 = torch.cat([x[:, :self.mid_idx], x[:, self.mid_idx + 1:]],
-                           dim=1).reshape(B * (N - 1), C, H, W)
-            f1 = mid.unsqueeze(1).expand(B, N - 1, C, H, W) \
-                .reshape(B * (N - 1), C, H, W)
-            fused = self.fuse(f0, f1).reshape(B, (N - 1) * C, H, W)
-            out = torch.cat([fused, mid], dim=1)
+            f0 = torch.cat([x[:m * B], x[(m + 1) * B:]], dim=0)
+            f1 = mid.repeat(N - 1, 1, 1, 1)
+            fused = self.fuse(f0, f1)                  # [(N-1)*B, C, H, W]
+            out = torch.cat([fused[i * B:(i + 1) * B] for i in range(N - 1)]
+                            + [mid], dim=1)
         else:
-            out = x.reshape(x.size(0), -1, x.size(-2), x.size(-1))
+            # channel concat of the N frames per batch item
+            out = torch.cat([x[i * B:(i + 1) * B] for i in range(N)], dim=1)
         return self.dense_fusion(out)
 
-    def scale_aggre(self, x, feats, idx):
+    def scale_aggre(self, x, feats, N, idx):
+        """feats FRAME-major [N*B, C, h, w]."""
         if self.has_scaleaggre:
-            B, N, C, H, W = feats.shape
-            flat = feats.reshape(B * N, C, H, W)
-            flat = flat * self.attens[idx](flat)
-            x = x + flat.reshape(B, N, C, H, W).mean(1)
+            B = feats.size(0) // N
+            flat = feats * self.attens[idx](feats)
+            x = x + flat.view(N, B, *flat.shape[1:]).mean(0)  # free view
         return self.recons[idx](x)
 
-    def forward(self, x, feats_list):
-        B, N = x.shape[:2]
+    def forward_frames(self, x, feats_list, N):
         assert N == self.num_frame
-        out = self.dense_fuse(x)
+        out = self.dense_fuse_frames(x, N)
         for idx, feats in enumerate(feats_list):
-            feats = feats.reshape(B, N, -1, feats.size(-2), feats.size(-1))
-            out = self.scale_aggre(out, feats, idx)
+            out = self.scale_aggre(out, feats, N, idx)
         return out
+
+    def forward(self, x, feats_list):
+        """5D-compat entry: x [B, N, C, H, W]; feats_list entries
+        [B*N, c, h, w] batch-major (as the encoder produces for a
+        batch-major flatten)."""
+        B, N = x.shape[:2]
+        xf = x.transpose(0, 1).reshape(N * B, *x.shape[2:])
+        ff = [f.view(B, N, *f.shape[1:]).transpose(0, 1)
+              .reshape(N * B, *f.shape[1:]) for f in feats_list]
+        return self.forward_frames(xf, ff, N)
 
 
 @register_model("ESRNet")
@@ -317,12 +329,13 @@ class ESRNet(nn.Module):
     def forward(self, x):
         B, N, C, H, W = x.shape
         x, box = _pad_multiple(x, self.DOWN_SCALE)
-        x = x.reshape(B * N, C, x.size(-2), x.size(-1))
+        # frame-major layout [N*B, C, H, W]: every temporal op below is a
+        # contiguous batch-slice concatenation — no 5D round trips, no
+        # index tensors (hipGraph-capture-safe), channels_last-compatible
+        x = x.transpose(0, 1).reshape(N * B, C, x.size(-2), x.size(-1))
         x = self.head(x)
         feats_list = self.feat_extract(x)
-        deep = feats_list[0]
-        deep = deep.reshape(B, N, -1, deep.size(-2), deep.size(-1))
-        deep = self.time_propagate(deep)
-        out = self.spacetime_fuse(deep, feats_list)
+        deep = self.time_propagate.forward_frames(feats_list[0], N)
+        out = self.spacetime_fuse.forward_frames(deep, feats_list, N)
         out = self.tail(out)
         return _crop(out, box)
