@@ -96,6 +96,12 @@ class Trainer:
         self.mid_idx = (train_dataloader.seqn - 1) // 2
         self.start_iteration = 0
 
+        # hipGraph-captured training step (GPU, fixed shapes):
+        # built lazily on the first batch
+        self.use_graphs = (tcfg.get("hip_graphs", False)
+                           and self.device.type == "cuda")
+        self._graph_step = None
+
         if resume is not None:
             self._resume_checkpoint(resume, reset)
 
@@ -110,10 +116,44 @@ class Trainer:
         import contextlib
         return contextlib.nullcontext()
 
+    def _get_graph_step(self, inputs_seq):
+        """Build (or fetch) the captured BPTT step for this window shape."""
+        from ..parallel import get_world_size
+        from .graph_runner import GraphedBPTTStep, flatten_grads
+        inp0 = inputs_seq[0]["inp_scaled_cnt"]
+        gt0 = inputs_seq[0]["gt_cnt"][:, self.mid_idx]
+        key = (len(inputs_seq), tuple(inp0.shape), tuple(gt0.shape))
+        if self._graph_step is not None and self._graph_step[0] == key:
+            return self._graph_step[1]
+        params = [p for p in self.model.parameters() if p.requires_grad]
+        flat = flatten_grads(params, self.device)
+        runner = GraphedBPTTStep(
+            self.model, self.optimizer, flat, len(inputs_seq),
+            tuple(inp0.shape), tuple(gt0.shape), self.device,
+            amp_dtype=self.amp_dtype,
+            world_size=get_world_size()).capture()
+        self._graph_step = (key, runner)
+        return runner
+
+    def graphed_bptt_step(self, inputs_seq):
+        runner = self._get_graph_step(inputs_seq)
+        loss, mse = runner.run(
+            [w["inp_scaled_cnt"] for w in inputs_seq],
+            [w["gt_cnt"][:, self.mid_idx] for w in inputs_seq])
+        return loss.detach().clone(), mse.detach().clone(), None
+
     def bptt_step(self, inputs_seq, train: bool = True):
         """One optimizer step over a window sequence: loss summed over the
         seqn-sliding windows, single backward through persistent GRU state
         (parity: ESR:train_ours_cnt_seq.py:210-235)."""
+        if train and self.use_graphs:
+            try:
+                return self.graphed_bptt_step(inputs_seq)
+            except Exception as e:
+                self.logger.warning(
+                    f"hipGraph step failed ({type(e).__name__}: {e}); "
+                    f"falling back to eager")
+                self.use_graphs = False
         if train:
             self.optimizer.zero_grad(set_to_none=True)
         self._unwrapped().reset_states()
@@ -262,7 +302,7 @@ class Trainer:
         ESR:train_ours_cnt_seq.py:258-293).  No-ops unless a TensorBoard
         sink is active."""
         vis_cfg = self.cfg["trainer"].get("vis", {"enabled": False})
-        if not vis_cfg.get("enabled", False):
+        if not vis_cfg.get("enabled", False) or pred is None:
             return
         if self.writer is None or getattr(self.writer, "_tb", None) is None:
             return
@@ -366,13 +406,24 @@ def build_training(config_parser, device, logger, resume=None, reset=False):
         if cfg.get("valid_dataloader") else None
 
     model = build_model(cfg["model"]["name"], **cfg["model"]["args"]).to(device)
-    model = wrap_ddp(model, device=device if device.type == "cuda" else None,
-                     sync_bn=cfg.get("sync_bn", False))
+    use_graphs = cfg["trainer"].get("hip_graphs", False) \
+        and device.type == "cuda"
+    if use_graphs:
+        # graphed mode does its own single-bucket RCCL all-reduce inside the
+        # captured step; DDP's hook-driven reducer is not used
+        if cfg.get("sync_bn", False):
+            raise ValueError("hip_graphs is incompatible with sync_bn")
+    else:
+        model = wrap_ddp(model, device=device if device.type == "cuda" else None,
+                         sync_bn=cfg.get("sync_bn", False))
 
     loss_fns = {"mse": nn.MSELoss(), "l1": nn.L1Loss()}
     params = [p for p in model.parameters() if p.requires_grad]
+    opt_kwargs = dict(cfg["optimizer"]["args"])
+    if use_graphs and cfg["optimizer"]["name"] in ("Adam", "AdamW"):
+        opt_kwargs["capturable"] = True
     optimizer = build_optimizer(cfg["optimizer"]["name"], params,
-                                **cfg["optimizer"]["args"])
+                                **opt_kwargs)
     lr_scheduler = build_lr_scheduler(cfg["lr_scheduler"]["name"], optimizer,
                                       **cfg["lr_scheduler"]["args"])
     return Trainer(config_parser, train_loader, valid_loader, model, loss_fns,
