@@ -36,6 +36,10 @@ __all__ = [
     "event_formatting",
     "normalize_events",
     "scaled_count_encoding",
+    "events_to_mask",
+    "events_polarity_mask",
+    "event_conversion",
+    "event_restore",
 ]
 
 
@@ -317,3 +321,55 @@ def scaled_count_encoding(normalized_events: torch.Tensor, sensor_resolution,
     if mode == "events":
         return torch.stack([(xs * W).long().float(), (ys * H).long().float(), ts, ps])
     raise ValueError(f"unsupported mode {mode}")
+
+
+def events_to_mask(xs: torch.Tensor, ys: torch.Tensor, ps: torch.Tensor,
+                   sensor_size) -> torch.Tensor:
+    """Binary activity mask: |p| written (not accumulated) per pixel.
+
+    Parity: ESR:dataloader/encodings.py:307-331.
+    """
+    H, W = int(sensor_size[0]), int(sensor_size[1])
+    idx, w = _mask_and_flatten(xs, ys, ps.float().abs(), sensor_size)
+    mask = torch.zeros(H * W, dtype=torch.float32, device=xs.device)
+    mask.scatter_(0, idx, w)
+    return mask.view(H, W)
+
+
+def events_polarity_mask(ps: torch.Tensor) -> torch.Tensor:
+    """[N] polarities -> [N, 2] one-hot-ish polarity mask (ch0 pos, ch1
+    |neg|).  Parity: ESR:dataloader/encodings.py:334-345."""
+    pos = ps.clamp(min=0)
+    neg = -ps.clamp(max=0)
+    return torch.stack([pos, neg], dim=1)
+
+
+def event_conversion(event_list: torch.Tensor, time_bins: int, resolution,
+                     time_bins_voxel: int | None = None) -> dict:
+    """Batched padded event cloud [B, N, 4] (x, y, t, p) -> dict of count /
+    voxel / stack encodings.  Parity: ESR:dataloader/encodings.py:536-577
+    (vectorized per item; events are re-sorted by t)."""
+    if time_bins_voxel is None:
+        time_bins_voxel = time_bins
+    e_cnt, e_voxel, e_stack = [], [], []
+    for entry in event_list.detach():
+        order = torch.argsort(entry[:, 2], stable=True)
+        ev = entry[order]
+        xs, ys, ts, ps = ev[:, 0], ev[:, 1], ev[:, 2], ev[:, 3]
+        e_cnt.append(events_to_channels(xs, ys, ps, resolution))
+        e_voxel.append(events_to_voxel(xs, ys, ts, ps, time_bins_voxel,
+                                       resolution))
+        e_stack.append(events_to_stack_no_polarity(xs, ys, ts, ps, time_bins,
+                                                   resolution))
+    return {"e_cnt": torch.stack(e_cnt), "e_voxel": torch.stack(e_voxel),
+            "e_stack": torch.stack(e_stack)}
+
+
+def event_restore(events: torch.Tensor, resolution) -> torch.Tensor:
+    """Normalized event cloud [B, N, 4] -> pixel coordinates with polarity
+    snapped to {-1, +1}.  Parity: ESR:dataloader/encodings.py:580-601."""
+    ev = events.detach().clone()
+    ev[:, :, 0] *= resolution[1]
+    ev[:, :, 1] *= resolution[0]
+    ev[:, :, 3] = torch.sign(ev[:, :, 3])
+    return ev

@@ -58,3 +58,27 @@ class EventVisualizer:
         except Exception:
             # fall back to raw npy dump if matplotlib is unavailable
             np.save(path + ".npy", img)
+
+
+def plot_event_stack(stack_bhw, is_save=False, path=None):
+    """Render a signed event stack [TB, H, W] as a grid of count images
+    (parity in spirit with ESR:myutils/vis_events/matplotlib_plot_events.py
+    stack plots)."""
+    import numpy as np
+    stack = np.asarray(stack_bhw, dtype=np.float64)
+    TB, H, W = stack.shape
+    viz = EventVisualizer()
+    tiles = []
+    for b in range(TB):
+        pos = np.clip(stack[b], 0, None)
+        neg = np.clip(-stack[b], 0, None)
+        tiles.append(viz.plot_event_cnt(np.stack([pos, neg], axis=-1)))
+    cols = int(np.ceil(np.sqrt(TB)))
+    rows = int(np.ceil(TB / cols))
+    canvas = np.zeros((rows * H, cols * W, 3), dtype=np.uint8)
+    for i, t in enumerate(tiles):
+        r, c = divmod(i, cols)
+        canvas[r * H:(r + 1) * H, c * W:(c + 1) * W] = t
+    if is_save and path:
+        EventVisualizer._save(canvas, path)
+    return canvas
