@@ -103,3 +103,33 @@ def test_distributed_sampler_partitions():
     idx0, idx1 = set(res[0][0]), set(res[1][0])
     assert len(idx0) == len(idx1) == 5
     assert idx0.isdisjoint(idx1)
+
+
+def _run_watchdog(rank, port, q):
+    try:
+        _init(rank, port)
+        from esr_amd.parallel.watchdog import Watchdog
+        events = []
+        wd = Watchdog(timeout=2.0, interval=0.2,
+                      on_failure=lambda ranks: events.append(ranks))
+        wd.beat()
+        import time
+        if rank == 0:
+            # rank 0 keeps beating; rank 1 goes silent after one beat
+            wd.start()
+            time.sleep(4.0)
+            wd.stop()
+            q.put((rank, list(wd.failed_ranks), len(events)))
+        else:
+            time.sleep(4.5)   # silent: no further beats
+            q.put((rank, [], 0))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(400)
+def test_watchdog_detects_silent_rank():
+    res = _spawn(_run_watchdog, 29517)
+    failed, n_events = res[0]
+    assert failed == [1], f"rank 0 should flag rank 1, got {failed}"
+    assert n_events >= 1
