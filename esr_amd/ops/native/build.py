@@ -18,7 +18,8 @@ from pathlib import Path
 HERE = Path(__file__).resolve().parent
 SRC = HERE / "src"
 SOURCES = [SRC / "bindings.cpp", SRC / "deform_conv.hip",
-           SRC / "gru_gates.hip", SRC / "event_ops.hip"]
+           SRC / "deform_conv_fused.hip", SRC / "gru_gates.hip",
+           SRC / "event_ops.hip"]
 
 
 def build(verbose: bool = False) -> str:

@@ -10,6 +10,9 @@ std::vector<at::Tensor> deform_conv2d_backward(
     const at::Tensor&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, const at::Tensor&,
     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t);
+at::Tensor deform_conv2d_forward_fused(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const c10::optional<at::Tensor>&, int64_t);
 at::Tensor deform_im2col_debug(
     const at::Tensor&, const at::Tensor&, const at::Tensor&,
     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
@@ -40,6 +43,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "modulated deformable conv backward (gfx950, batched)");
   m.def("deform_im2col", &deform_im2col_debug,
         "deformable im2col (test hook)");
+  m.def("deform_conv2d_forward_fused", &deform_conv2d_forward_fused,
+        "fused im2col+MFMA deformable conv forward (test hook)");
   m.def("gru_gates_ur_forward", &gru_gates_ur_forward);
   m.def("gru_gates_ur_backward", &gru_gates_ur_backward);
   m.def("gru_gates_out_forward", &gru_gates_out_forward);

@@ -238,6 +238,14 @@ at::Tensor dcn_im2col(const at::Tensor& input, const at::Tensor& offset,
 
 }  // namespace
 
+// deform_conv_fused.hip
+bool dcn_fused_applicable(const at::Tensor&, const at::Tensor&, int64_t,
+                          int64_t, int64_t, int64_t, int64_t, int64_t,
+                          int64_t);
+at::Tensor deform_conv2d_forward_fused(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const c10::optional<at::Tensor>&, int64_t);
+
 at::Tensor deform_conv2d_forward(
     const at::Tensor& input, const at::Tensor& offset, const at::Tensor& mask,
     const at::Tensor& weight, const c10::optional<at::Tensor>& bias,
@@ -251,6 +259,9 @@ at::Tensor deform_conv2d_forward(
   TORCH_CHECK(offset.size(1) == g.dg * 2 * g.kh * g.kw, "offset channels");
   TORCH_CHECK(mask.size(1) == g.dg * g.kh * g.kw, "mask channels");
   TORCH_CHECK(g.C % g.dg == 0, "C % deformable_groups != 0");
+
+  if (dcn_fused_applicable(input, weight, sh, sw, ph, pw, dh, dw, dg))
+    return deform_conv2d_forward_fused(input, offset, mask, weight, bias, dg);
 
   auto cols = dcn_im2col(input, offset, mask, g);
   // one batched GEMM: [B, Cout, C*K] x [B, C*K, HoWo]

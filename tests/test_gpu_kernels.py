@@ -261,3 +261,45 @@ class TestBenchPath:
             idx = c * 64 + ev[:, 1].long() * 8 + ev[:, 0].long()
             rebuilt.view(-1).scatter_add_(0, idx, ev[:, 3])
             assert torch.allclose(rebuilt, stack[b])
+
+
+class TestFusedDCN:
+    """Fused im2col+MFMA forward vs the generic im2col+GEMM path."""
+
+    @pytest.mark.parametrize("B,C,H,W,Cout,dg", [
+        (2, 16, 16, 16, 16, 4),
+        (4, 64, 32, 32, 64, 8),     # the ESRNet alignment shape
+        (1, 32, 17, 23, 48, 2),     # odd spatial size
+    ])
+    def test_fused_matches_generic(self, ext, B, C, H, W, Cout, dg):
+        g = torch.Generator().manual_seed(42)
+        input = torch.randn(B, C, H, W, generator=g).cuda()
+        offset = (torch.randn(B, dg * 18, H, W, generator=g) * 2).cuda()
+        mask = torch.rand(B, dg * 9, H, W, generator=g).cuda()
+        weight = (torch.randn(Cout, C, 3, 3, generator=g) * 0.2).cuda()
+        bias = torch.randn(Cout, generator=g).cuda()
+        fused = ext.deform_conv2d_forward_fused(input, offset, mask, weight,
+                                                bias, dg)
+        # generic = im2col + GEMM (itself oracle-verified)
+        cols = ext.deform_im2col(input, offset, mask, 3, 3, 1, 1, 1, 1,
+                                 1, 1, dg)
+        ref = torch.matmul(weight.reshape(Cout, -1), cols) \
+            .reshape(B, Cout, H, W) + bias.view(1, -1, 1, 1)
+        err = (fused - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err / scale < 1e-5, f"rel err {err/scale}"
+
+    def test_dispatch_uses_fused_for_model_shape(self, ext):
+        # model shape must satisfy the fused-path predicate; verified by
+        # numeric equality of the public entry with the fused test hook
+        g = torch.Generator().manual_seed(1)
+        input = torch.randn(2, 64, 32, 32, generator=g).cuda()
+        offset = (torch.randn(2, 8 * 18, 32, 32, generator=g)).cuda()
+        mask = torch.rand(2, 8 * 9, 32, 32, generator=g).cuda()
+        weight = (torch.randn(64, 64, 3, 3, generator=g) * 0.2).cuda()
+        bias = torch.randn(64, generator=g).cuda()
+        pub = ext.deform_conv2d_forward(input, offset, mask, weight, bias,
+                                        1, 1, 1, 1, 1, 1, 8)
+        hook = ext.deform_conv2d_forward_fused(input, offset, mask, weight,
+                                               bias, 8)
+        assert torch.equal(pub, hook)
