@@ -1,4 +1,11 @@
+import os
+
 import pytest
+
+# Tests are correctness runs: skip MIOpen's exhaustive kernel auto-tuning
+# (its bf16 naive-probe kernels cost minutes per new conv-shape set).  The
+# driver's official bench.py runs are NOT under pytest and keep full find.
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
 
 
 def pytest_configure(config):

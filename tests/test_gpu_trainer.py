@@ -10,14 +10,14 @@ import torch
 pytestmark = pytest.mark.gpu
 
 
-def _train_config(datalist, out_dir, iterations=4, hip_graphs=True):
+def _train_config(datalist, out_dir, iterations=2, hip_graphs=True):
     ds = {
         "scale": 2, "ori_scale": "down4", "time_bins": 1,
         "need_gt_frame": False, "need_gt_events": True,
         "mode": "events", "window": 1024, "sliding_window": 512,
         "data_augment": {"enabled": False, "augment": [], "augment_prob": []},
         "hot_filter": {"enabled": False},
-        "sequence": {"sequence_length": 4, "seqn": 3, "step_size": None,
+        "sequence": {"sequence_length": 3, "seqn": 3, "step_size": None,
                      "pause": {"enabled": False,
                                "proba_pause_when_running": 0.05,
                                "proba_pause_when_paused": 0.9}},
@@ -89,7 +89,7 @@ def test_trainer_eager_vs_graph_losses_close(tmp_path, gpu_synth_datalist):
                                hip_graphs=False, run_id="e1")
     torch.manual_seed(7)
     _, t_graph = _run_training(tmp_path, gpu_synth_datalist,
-                               hip_graphs=True, run_id="g2")
+                               hip_graphs=True, run_id="g2")  # reuses find db
     a = t_eager.train_metrics.avg("train_mse_loss")
     b = t_graph.train_metrics.avg("train_mse_loss")
     assert abs(a - b) / max(abs(a), 1e-6) < 0.1, (a, b)
