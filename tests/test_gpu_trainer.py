@@ -37,7 +37,7 @@ def _train_config(datalist, out_dir, iterations=2, hip_graphs=True):
             "hip_graphs": hip_graphs,
             "epoch_based_train": {"enabled": False},
             "iteration_based_train": {
-                "enabled": True, "iterations": iterations, "save_period": 2,
+                "enabled": True, "iterations": iterations, "save_period": 1,
                 "train_log_step": 1, "valid_log_step": 1, "valid_step": 100,
                 "lr_change_rate": 1000},
             "monitor": "off", "tensorboard": False,
