@@ -97,6 +97,9 @@ class StreamingESR:
             # run once eagerly to materialize the state shape
             out = self._forward(self._static_in)
             self._static_state = holder.state.detach().clone()
+            # warmup and capture both EXECUTE the body, each advancing the
+            # in-place state buffer — snapshot and restore the live state
+            state_snapshot = self._static_state.clone()
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(s):
@@ -108,6 +111,8 @@ class StreamingESR:
             with torch.cuda.graph(g):
                 self._static_out = self._graph_body()
             torch.cuda.synchronize()
+            self._static_state.copy_(state_snapshot)
+            holder.state = self._static_state
             self._graph = g
             return out
         self._static_in.copy_(inp)
