@@ -128,7 +128,9 @@ class StreamingESR:
         inp = torch.stack(tuple(self._frames))[None]     # [1, seqn, 2, kH, kW]
         if self.use_graphs:
             try:
-                return self._forward_graphed(inp)[0]
+                # clone: the graph writes into one static output buffer,
+                # the caller gets an independent tensor
+                return self._forward_graphed(inp)[0].clone()
             except Exception:
                 self.use_graphs = False
         return self._forward(inp)[0]
