@@ -30,41 +30,41 @@ import torch.distributed as dist
 import torch.nn.functional as F
 
 
-def make_window_pool(n_sets, n_windows, batch, seqn, window, lr_res, hr_res,
-                     device, ext, seed):
-    """Generate pools of (inp_scaled_cnt [B,seqn,2,kH,kW], gt_cnt [B,2,kH,kW])
-    per window, fully on device via the native splat kernels."""
+def make_sequence_pool(n_sets, seql, n_windows, batch, seqn, window, lr_res,
+                       hr_res, device, ext, seed):
+    """Generate pools of (frames [B,seql,2,kH,kW], gts: n_windows x
+    [B,2,kH,kW]) per step, fully on device via the native splat kernels.
+    A BPTT step has seql UNIQUE frames; the seqn-windows are views."""
     g = torch.Generator(device=device).manual_seed(seed)
     H, W = lr_res
     kH, kW = hr_res
     scale = kH // H
     pools = []
     for _ in range(n_sets):
-        windows = []
+        BF = batch * seql
+        ev = torch.empty(BF, window, 4, device=device)
+        ev[..., 0] = (torch.rand(BF, window, device=device, generator=g)
+                      * W).floor() * scale
+        ev[..., 1] = (torch.rand(BF, window, device=device, generator=g)
+                      * H).floor() * scale
+        ev[..., 2] = torch.rand(BF, window, device=device, generator=g)
+        ev[..., 3] = torch.randint(0, 2, (BF, window), device=device,
+                                   generator=g).float() * 2 - 1
+        frames = ext.splat_count(ev.contiguous(), kH, kW) \
+            .view(batch, seql, 2, kH, kW)
+        gts = []
+        ng = window * scale * scale
         for _ in range(n_windows):
-            BF = batch * seqn
-            ev = torch.empty(BF, window, 4, device=device)
-            ev[..., 0] = (torch.rand(BF, window, device=device, generator=g)
-                          * W).floor() * scale
-            ev[..., 1] = (torch.rand(BF, window, device=device, generator=g)
-                          * H).floor() * scale
-            ev[..., 2] = torch.rand(BF, window, device=device, generator=g)
-            ev[..., 3] = torch.randint(0, 2, (BF, window), device=device,
-                                       generator=g).float() * 2 - 1
-            inp = ext.splat_count(ev.contiguous(), kH, kW) \
-                .view(batch, seqn, 2, kH, kW)
-            gt_ev = torch.empty(batch, window * scale * scale, 4, device=device)
-            gt_ev[..., 0] = (torch.rand(batch, window * scale * scale,
-                                        device=device, generator=g) * kW).floor()
-            gt_ev[..., 1] = (torch.rand(batch, window * scale * scale,
-                                        device=device, generator=g) * kH).floor()
-            gt_ev[..., 2] = torch.rand(batch, window * scale * scale,
-                                       device=device, generator=g)
-            gt_ev[..., 3] = torch.randint(0, 2, (batch, window * scale * scale),
-                                          device=device, generator=g).float() * 2 - 1
-            gt = ext.splat_count(gt_ev.contiguous(), kH, kW)
-            windows.append((inp, gt))
-        pools.append(windows)
+            gt_ev = torch.empty(batch, ng, 4, device=device)
+            gt_ev[..., 0] = (torch.rand(batch, ng, device=device,
+                                        generator=g) * kW).floor()
+            gt_ev[..., 1] = (torch.rand(batch, ng, device=device,
+                                        generator=g) * kH).floor()
+            gt_ev[..., 2] = torch.rand(batch, ng, device=device, generator=g)
+            gt_ev[..., 3] = torch.randint(0, 2, (batch, ng), device=device,
+                                          generator=g).float() * 2 - 1
+            gts.append(ext.splat_count(gt_ev.contiguous(), kH, kW))
+        pools.append((frames, gts))
     return pools
 
 
@@ -149,14 +149,15 @@ def main():
     lr_res = (args.lr_size, args.lr_size)
     hr_res = (args.lr_size * args.scale, args.lr_size * args.scale)
     n_windows = args.seql - args.seqn + 1
-    pools = make_window_pool(2, n_windows, args.batch, args.seqn, args.window,
-                             lr_res, hr_res, device, ext, seed=100 + rank)
+    pools = make_sequence_pool(2, args.seql, n_windows, args.batch,
+                               args.seqn, args.window, lr_res, hr_res,
+                               device, ext, seed=100 + rank)
     amp_dtype = torch.bfloat16 if args.dtype == "bf16" else None
 
     # graph replay reads from fixed addresses: one graph per pre-generated
     # data set (no per-step copies), all sharing one memory pool
-    static_in = [inp for inp, _ in pools[0]]
-    static_gt = [gt for _, gt in pools[0]]
+    static_frames = [pools[0][0]]
+    static_gts = [list(pools[0][1])]
 
     import contextlib
 
@@ -169,12 +170,17 @@ def main():
 
     def fwd_bwd():
         flat_grad.zero_()
-        model.reset_states()
+        inner = model.module if hasattr(model, "module") else model
+        inner.reset_states()
+        frames = static_frames[0]
+        x = frames.to(torch.bfloat16) if pure_bf16 else frames
+        # shared-encoder BPTT: head/encoder run once per UNIQUE frame, the
+        # seqn-windows are contiguous frame-major slices (math-identical to
+        # the per-window loop — tests/test_model.py::test_forward_sequence_*)
+        with autocast():
+            preds = inner.forward_sequence(x, args.seqn)
         loss = 0
-        for inp, gt in zip(static_in, static_gt):
-            x = inp.to(torch.bfloat16) if pure_bf16 else inp
-            with autocast():
-                pred = model(x)
+        for pred, gt in zip(preds, static_gts[0]):
             loss = loss + F.mse_loss(pred.float(), gt)
         loss.backward()
         return loss
@@ -194,10 +200,9 @@ def main():
             optimizer.step()
 
     def set_data(i):
-        windows = pools[i % len(pools)]
-        for w, (inp, gt) in enumerate(windows):
-            static_in[w] = inp
-            static_gt[w] = gt
+        frames, gts = pools[i % len(pools)]
+        static_frames[0] = frames
+        static_gts[0] = list(gts)
 
     def _warmup_side_stream(body):
         s = torch.cuda.Stream()

@@ -339,3 +339,30 @@ class ESRNet(nn.Module):
         out = self.spacetime_fuse.forward_frames(deep, feats_list, N)
         out = self.tail(out)
         return _crop(out, box)
+
+    def forward_sequence(self, frames, seqn: int):
+        """BPTT over all sliding seqn-windows of a sequence with the
+        head/encoder computed ONCE per unique frame.
+
+        frames: [B, L, C, H, W] (L >= seqn).  Returns the list of
+        L-seqn+1 outputs [B, inch, H, W] — mathematically identical to
+        calling forward() per window (the reference's loop,
+        ESR:train_ours_cnt_seq.py:217-232): per-frame encoders are
+        frame-local, so overlapping windows share their features; autograd
+        accumulates every window's gradient into the shared encoder pass.
+        In frame-major layout each window's features are a contiguous
+        slice — no gather, no copy.
+        """
+        B, L, C, H, W = frames.shape
+        x, box = _pad_multiple(frames, self.DOWN_SCALE)
+        x = x.transpose(0, 1).reshape(L * B, C, x.size(-2), x.size(-1))
+        x = self.head(x)
+        feats_list = self.feat_extract(x)
+        outs = []
+        for w in range(L - seqn + 1):
+            sl = slice(w * B, (w + seqn) * B)
+            deep = self.time_propagate.forward_frames(feats_list[0][sl], seqn)
+            out = self.spacetime_fuse.forward_frames(
+                deep, [f[sl] for f in feats_list], seqn)
+            outs.append(_crop(self.tail(out), box))
+        return outs

@@ -85,3 +85,31 @@ def test_num_frame_5():
     m = build_model("ESRNet", inch=2, basech=8, num_frame=5)
     y = m(torch.randn(1, 5, 2, 16, 16))
     assert y.shape == (1, 2, 16, 16)
+
+
+def test_forward_sequence_matches_per_window():
+    """forward_sequence (shared encoder) must equal per-window forward()
+    exactly, including gradient totals."""
+    torch.manual_seed(3)
+    m = build_model("ESRNet", inch=2, basech=8, num_frame=3)
+    frames = torch.rand(2, 5, 2, 32, 32)
+
+    m.reset_states()
+    outs_seq = m.forward_sequence(frames, seqn=3)
+    loss_seq = sum((o ** 2).mean() for o in outs_seq)
+    loss_seq.backward()
+    grads_seq = [p.grad.clone() for p in m.parameters() if p.grad is not None]
+    for p in m.parameters():
+        p.grad = None
+
+    m.reset_states()
+    outs_ref = [m(frames[:, w:w + 3]) for w in range(3)]
+    loss_ref = sum((o ** 2).mean() for o in outs_ref)
+    loss_ref.backward()
+    grads_ref = [p.grad.clone() for p in m.parameters() if p.grad is not None]
+
+    for a, b in zip(outs_seq, outs_ref):
+        assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max().item()
+    assert abs(loss_seq.item() - loss_ref.item()) < 1e-6
+    for ga, gb in zip(grads_seq, grads_ref):
+        assert torch.allclose(ga, gb, atol=1e-5), (ga - gb).abs().max().item()
