@@ -19,7 +19,7 @@ from .sequence import SequenceDataset
 
 __all__ = ["read_datalist", "concatenate_datasets", "make_event_loader",
            "SequenceDataLoader", "InferenceSequenceDataLoader",
-           "sequence_collate"]
+           "sequence_collate", "shared_sequence_collate"]
 
 _PACK_KEYS = {"inp_events", "inp_normalized_events", "inp_scaled_events",
               "inp_pol_mask", "gt_events", "gt_normalized_events"}
@@ -109,11 +109,15 @@ def make_event_loader(dataloader_config):
 def SequenceDataLoader(dataloader_config):
     """Sequence loader over a datalist (parity:
     ESR:dataloader/h5dataloader.py:180-208).  Returns a DataLoader with
-    .seqn/.dist_sampler/.inp_sensor_resolution/.gt_sensor_resolution set."""
+    .seqn/.dist_sampler/.inp_sensor_resolution/.gt_sensor_resolution set.
+    collate='shared' yields shared-encoder sequence batches instead of
+    materialized windows."""
     ds = concatenate_datasets(dataloader_config["path_to_datalist_txt"],
                               SequenceDataset, dataloader_config["dataset"])
     seqn = dataloader_config["dataset"]["sequence"]["seqn"]
-    loader, sampler = _make_loader(ds, dataloader_config, sequence_collate(seqn))
+    collate_cls = shared_sequence_collate \
+        if dataloader_config.get("collate") == "shared" else sequence_collate
+    loader, sampler = _make_loader(ds, dataloader_config, collate_cls(seqn))
     loader.seqn = seqn
     loader.dist_sampler = sampler
     loader.gt_sensor_resolution = ds.datasets[0].gt_sensor_resolution
@@ -136,3 +140,28 @@ def InferenceSequenceDataLoader(data_path, dataloader_config):
     loader.inp_sensor_resolution = ds.inp_sensor_resolution
     loader.scale = dataloader_config["dataset"]["scale"]
     return loader
+
+
+class shared_sequence_collate:
+    """Collate for shared-encoder BPTT (ESRNet.forward_sequence): returns
+    one dict per batch with the L UNIQUE window frames plus the per-window
+    mid-frame ground truths, instead of L-seqn+1 materialized windows."""
+
+    def __init__(self, seqn: int):
+        self.seqn = seqn
+        self.mid_idx = (seqn - 1) // 2
+
+    def __call__(self, batch: list[list[dict]]) -> dict:
+        L = len(batch[0])
+        per_step = [_stack_items([entry[i] for entry in batch])
+                    for i in range(L)]
+        assert L >= self.seqn
+        n_windows = L - self.seqn + 1
+        frames = torch.stack([s["inp_scaled_cnt"] for s in per_step], dim=1)
+        gt_mids = torch.stack(
+            [per_step[w + self.mid_idx]["gt_cnt"] for w in range(n_windows)],
+            dim=1)
+        return {"frames": frames,            # [B, L, 2, kH, kW]
+                "gt_mids": gt_mids,          # [B, n_windows, 2, kH, kW]
+                "seqn": self.seqn,
+                "mid_step": per_step[self.mid_idx]}

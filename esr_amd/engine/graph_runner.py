@@ -39,17 +39,25 @@ class GraphedBPTTStep:
 
     def __init__(self, model, optimizer, flat_grad, n_windows,
                  inp_shape, gt_shape, device, amp_dtype=None,
-                 world_size: int = 1, warmup: int = 3):
+                 world_size: int = 1, warmup: int = 3,
+                 sequence: bool = False, seqn: int = 3):
         self.model = model
         self.optimizer = optimizer
         self.flat_grad = flat_grad
         self.device = device
         self.amp_dtype = amp_dtype
         self.world = world_size
-        self.static_in = [torch.zeros(inp_shape, device=device)
-                          for _ in range(n_windows)]
-        self.static_gt = [torch.zeros(gt_shape, device=device)
-                          for _ in range(n_windows)]
+        self.sequence = sequence
+        self.seqn = seqn
+        if sequence:
+            # inp_shape = [B, L, C, H, W]; gt_shape = [B, n_windows, C, H, W]
+            self.static_in = [torch.zeros(inp_shape, device=device)]
+            self.static_gt = [torch.zeros(gt_shape, device=device)]
+        else:
+            self.static_in = [torch.zeros(inp_shape, device=device)
+                              for _ in range(n_windows)]
+            self.static_gt = [torch.zeros(gt_shape, device=device)
+                              for _ in range(n_windows)]
         self.graph = None
         self.static_loss = None
         self._warmup = warmup
@@ -67,11 +75,19 @@ class GraphedBPTTStep:
         inner.reset_states()
         loss = 0
         mse = None
-        for inp, gt in zip(self.static_in, self.static_gt):
+        if self.sequence:
             with self._autocast():
-                pred = self.model(inp)
-            mse = F.mse_loss(pred.float(), gt)
-            loss = loss + mse
+                preds = inner.forward_sequence(self.static_in[0], self.seqn)
+            gts = self.static_gt[0]
+            for w, pred in enumerate(preds):
+                mse = F.mse_loss(pred.float(), gts[:, w])
+                loss = loss + mse
+        else:
+            for inp, gt in zip(self.static_in, self.static_gt):
+                with self._autocast():
+                    pred = self.model(inp)
+                mse = F.mse_loss(pred.float(), gt)
+                loss = loss + mse
         loss.backward()
         if self.world > 1:
             dist.all_reduce(self.flat_grad)

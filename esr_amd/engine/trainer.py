@@ -117,29 +117,40 @@ class Trainer:
         return contextlib.nullcontext()
 
     def _get_graph_step(self, inputs_seq):
-        """Build (or fetch) the captured BPTT step for this window shape."""
+        """Build (or fetch) the captured BPTT step for this batch shape."""
         from ..parallel import get_world_size
         from .graph_runner import GraphedBPTTStep, flatten_grads
-        inp0 = inputs_seq[0]["inp_scaled_cnt"]
-        gt0 = inputs_seq[0]["gt_cnt"][:, self.mid_idx]
-        key = (len(inputs_seq), tuple(inp0.shape), tuple(gt0.shape))
+        shared = isinstance(inputs_seq, dict)
+        if shared:
+            inp_shape = tuple(inputs_seq["frames"].shape)
+            gt_shape = tuple(inputs_seq["gt_mids"].shape)
+            n_windows = gt_shape[1]
+        else:
+            inp_shape = tuple(inputs_seq[0]["inp_scaled_cnt"].shape)
+            gt_shape = tuple(inputs_seq[0]["gt_cnt"][:, self.mid_idx].shape)
+            n_windows = len(inputs_seq)
+        key = (shared, n_windows, inp_shape, gt_shape)
         if self._graph_step is not None and self._graph_step[0] == key:
             return self._graph_step[1]
         params = [p for p in self.model.parameters() if p.requires_grad]
         flat = flatten_grads(params, self.device)
         runner = GraphedBPTTStep(
-            self.model, self.optimizer, flat, len(inputs_seq),
-            tuple(inp0.shape), tuple(gt0.shape), self.device,
-            amp_dtype=self.amp_dtype,
-            world_size=get_world_size()).capture()
+            self.model, self.optimizer, flat, n_windows,
+            inp_shape, gt_shape, self.device,
+            amp_dtype=self.amp_dtype, world_size=get_world_size(),
+            sequence=shared, seqn=self.train_dataloader.seqn).capture()
         self._graph_step = (key, runner)
         return runner
 
     def graphed_bptt_step(self, inputs_seq):
         runner = self._get_graph_step(inputs_seq)
-        loss, mse = runner.run(
-            [w["inp_scaled_cnt"] for w in inputs_seq],
-            [w["gt_cnt"][:, self.mid_idx] for w in inputs_seq])
+        if isinstance(inputs_seq, dict):
+            loss, mse = runner.run([inputs_seq["frames"]],
+                                   [inputs_seq["gt_mids"]])
+        else:
+            loss, mse = runner.run(
+                [w["inp_scaled_cnt"] for w in inputs_seq],
+                [w["gt_cnt"][:, self.mid_idx] for w in inputs_seq])
         return loss.detach().clone(), mse.detach().clone(), None
 
     def bptt_step(self, inputs_seq, train: bool = True):
@@ -159,17 +170,36 @@ class Trainer:
         self._unwrapped().reset_states()
         loss = 0
         mse_loss = None
-        for inputs in inputs_seq:
-            inp = inputs["inp_scaled_cnt"].to(self.device, non_blocking=True)
-            gt = inputs["gt_cnt"][:, self.mid_idx].to(self.device,
-                                                      non_blocking=True)
+        if isinstance(inputs_seq, dict):
+            # shared-encoder sequence batch (collate='shared'); bypasses the
+            # DDP wrapper's forward, so it requires either single process or
+            # the graphed step's own all-reduce
+            if train and hasattr(self.model, "module"):
+                raise RuntimeError(
+                    "collate='shared' training under DDP requires "
+                    "trainer.hip_graphs (the graphed step all-reduces the "
+                    "flat gradient itself)")
+            frames = inputs_seq["frames"].to(self.device, non_blocking=True)
+            gts = inputs_seq["gt_mids"].to(self.device, non_blocking=True)
             with self._autocast():
-                pred = self.model(inp)
-                if pred.shape[-2:] != gt.shape[-2:]:
-                    pred = F.interpolate(pred, size=gt.shape[-2:],
-                                         mode="bicubic", align_corners=False)
-                mse_loss = self.loss_fns["mse"](pred.float(), gt.float())
-            loss = loss + mse_loss
+                preds = self._unwrapped().forward_sequence(
+                    frames, inputs_seq["seqn"])
+            for w, pred in enumerate(preds):
+                mse_loss = self.loss_fns["mse"](pred.float(), gts[:, w].float())
+                loss = loss + mse_loss
+            pred = preds[-1]
+        else:
+            for inputs in inputs_seq:
+                inp = inputs["inp_scaled_cnt"].to(self.device, non_blocking=True)
+                gt = inputs["gt_cnt"][:, self.mid_idx].to(self.device,
+                                                          non_blocking=True)
+                with self._autocast():
+                    pred = self.model(inp)
+                    if pred.shape[-2:] != gt.shape[-2:]:
+                        pred = F.interpolate(pred, size=gt.shape[-2:],
+                                             mode="bicubic", align_corners=False)
+                    mse_loss = self.loss_fns["mse"](pred.float(), gt.float())
+                loss = loss + mse_loss
         if train:
             loss.backward()
             self.optimizer.step()
@@ -309,6 +339,8 @@ class Trainer:
         step = vis_cfg.get("train_img_writer_num", 20)
         if iter_idx % step != 0:
             return
+        if isinstance(inputs_seq, dict):
+            return  # shared-sequence batches carry no per-window dicts
         from ..utils.vis import EventVisualizer
         vis = EventVisualizer()
         inputs = inputs_seq[-1]

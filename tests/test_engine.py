@@ -120,3 +120,17 @@ def test_monitor_early_stop_logic(trained):
     assert not stop and not best
     stop, best = trainer.eval_model_performance({"valid_loss": 2.0})
     assert stop
+
+
+def test_shared_collate_training(tmp_path, synth_datalist):
+    """collate='shared' trains through forward_sequence with the same
+    machinery (CPU, single process)."""
+    cfg = _train_config(synth_datalist, tmp_path / "out_shared", iterations=2)
+    cfg["train_dataloader"]["collate"] = "shared"
+    cfg["valid_dataloader"]["collate"] = "shared"
+    parser = ConfigParser(cfg, run_id="sh0")
+    logger = setup_logging("test-shared", None)
+    trainer = build_training(parser, torch.device("cpu"), logger)
+    trainer.train()
+    assert math.isfinite(trainer.train_metrics.avg("train_loss"))
+    assert trainer.train_metrics.avg("train_loss") > 0
