@@ -8,7 +8,6 @@ strings, and the parser is usable without creating run directories (tests).
 from __future__ import annotations
 
 import argparse
-import os
 import time
 from pathlib import Path
 

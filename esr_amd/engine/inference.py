@@ -11,7 +11,6 @@ the reference's host wall-clock.
 
 from __future__ import annotations
 
-import os
 import time
 from pathlib import Path
 

@@ -22,7 +22,7 @@ import torch
 import torch.nn.functional as F
 
 from ..config import build_lr_scheduler, build_optimizer
-from ..parallel import get_rank, is_distributed, reduce_tensor
+from ..parallel import get_rank, reduce_tensor
 from ..utils import MetricTracker, MetricWriter
 from .checkpoint import Resumer, save_checkpoint
 

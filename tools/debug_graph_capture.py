@@ -8,7 +8,6 @@ from pathlib import Path
 sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 import torch
-import torch.nn.functional as F
 
 from esr_amd.models import build_model
 from esr_amd.ops.native import require_ext
