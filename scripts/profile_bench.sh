@@ -14,7 +14,9 @@ export TMPDIR=/tmp
 cd /tmp
 timeout 400 rocprofv3 --kernel-trace --stats --output-format csv -d "$PROF" \
     -o "$NAME" -- bash -c "cd '$REPO' && python bench.py --steps 3 --warmup 2 $*"
-# keep CSVs only; the .db can be hundreds of MiB
+# keep CSVs only; the .db can be hundreds of MiB; gzip the trace (the
+# copy-back budget is 64 MiB total)
 find "$PROF" -name "${NAME}*" ! -name "*.csv" -delete || true
+gzip -f "$PROF/${NAME}_kernel_trace.csv" 2>/dev/null || true
 ls -la "$PROF"
 echo "profile written to gpurun_out/prof/${NAME}*"
