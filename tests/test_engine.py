@@ -134,3 +134,19 @@ def test_shared_collate_training(tmp_path, synth_datalist):
     trainer.train()
     assert math.isfinite(trainer.train_metrics.avg("train_loss"))
     assert trainer.train_metrics.avg("train_loss") > 0
+
+
+def test_epoch_based_training(tmp_path, synth_datalist):
+    cfg = _train_config(synth_datalist, tmp_path / "out_epoch")
+    cfg["trainer"]["iteration_based_train"] = {"enabled": False}
+    cfg["trainer"]["epoch_based_train"] = {
+        "enabled": True, "epochs": 2, "save_period": 1,
+        "train_log_step": 1, "valid_log_step": 1, "valid_step": 1}
+    parser = ConfigParser(cfg, run_id="ep0")
+    logger = setup_logging("test-epoch", None)
+    trainer = build_training(parser, torch.device("cpu"), logger)
+    trainer.train()
+    ckpts = list(Path(parser.save_dir).glob("checkpoint-epoch*.pth"))
+    assert ckpts, "no epoch checkpoint written"
+    state = torch.load(ckpts[-1], map_location="cpu", weights_only=False)
+    assert state["trainer"]["training_mode"] == "epoch_based_train"
