@@ -114,6 +114,92 @@ __global__ void dcn_im2col_kernel(
   }
 }
 
+// --------------------------------------------------- col2im (LDS-tiled)
+// grad wrt input for the 3x3/stride1/pad1 shape: one block per
+// (b, c, 16x16 output tile).  Corner contributions that land inside the
+// tile's input region (+/- HALO) accumulate via fast LDS atomics and are
+// flushed once per block (one global atomic per covered input pixel —
+// ~16x fewer HBM atomics than the per-contribution kernel below); corners
+// pushed outside by large learned offsets fall back to device-scope
+// global atomics, so correctness never depends on the offset magnitude.
+constexpr int C2I_TILE = 16;
+constexpr int C2I_HALO = 4;
+constexpr int C2I_EDGE = C2I_TILE + 2 * C2I_HALO;   // 24
+
+__global__ __launch_bounds__(256)
+void dcn_col2im_tiled_kernel(
+    const float* __restrict__ col_grad, const float* __restrict__ offset,
+    const float* __restrict__ mask, DcnGeom g, float* __restrict__ grad_im) {
+  const int K = g.kh * g.kw;            // 9
+  const int HoWo = g.Ho * g.Wo;
+  const int cpg = g.C / g.dg;
+
+  const int tiles_x = (g.Wo + C2I_TILE - 1) / C2I_TILE;
+  const int tiles_y = (g.Ho + C2I_TILE - 1) / C2I_TILE;
+  const int tile = blockIdx.x;
+  const int ty0 = (tile / tiles_x) * C2I_TILE;
+  const int tx0 = (tile % tiles_x) * C2I_TILE;
+  const int c = blockIdx.y;
+  const int b = blockIdx.z;
+  const int grp = c / cpg;
+
+  __shared__ float acc[C2I_EDGE * C2I_EDGE];
+  for (int i = threadIdx.x; i < C2I_EDGE * C2I_EDGE; i += 256)
+    acc[i] = 0.f;
+  __syncthreads();
+
+  const int px_local = threadIdx.x;                 // 256 = 16x16 tile
+  const int ho = ty0 + px_local / C2I_TILE;
+  const int wo = tx0 + px_local % C2I_TILE;
+  float* gim = grad_im + ((long long)b * g.C + c) * g.H * g.W;
+
+  if (ho < g.Ho && wo < g.Wo) {
+    const int pix = ho * g.Wo + wo;
+    const float* off_p = offset + ((long long)b * g.dg + grp) * 2 * K * HoWo;
+    const float* msk_p = mask + ((long long)b * g.dg + grp) * K * HoWo;
+    const float* cg_p = col_grad + (((long long)b * g.C + c) * K) * HoWo + pix;
+    #pragma unroll
+    for (int k = 0; k < 9; ++k) {
+      const int i = k / 3, j = k % 3;
+      const float off_h = off_p[(2 * k) * HoWo + pix];
+      const float off_w = off_p[(2 * k + 1) * HoWo + pix];
+      const float m = msk_p[k * HoWo + pix];
+      const float h_im = ho - 1 + i + off_h;
+      const float w_im = wo - 1 + j + off_w;
+      if (h_im <= -1 || w_im <= -1 || h_im >= g.H || w_im >= g.W) continue;
+      const float gval = cg_p[(long long)k * HoWo] * m;
+      const int h0 = (int)floorf(h_im);
+      const int w0 = (int)floorf(w_im);
+      const float lh = h_im - h0, lw = w_im - w0;
+      const float wgt[4] = {(1 - lh) * (1 - lw), (1 - lh) * lw,
+                            lh * (1 - lw), lh * lw};
+      #pragma unroll
+      for (int corner = 0; corner < 4; ++corner) {
+        const int hh = h0 + (corner >> 1);
+        const int ww = w0 + (corner & 1);
+        if (hh < 0 || hh >= g.H || ww < 0 || ww >= g.W) continue;
+        const int ly = hh - (ty0 - C2I_HALO);
+        const int lx = ww - (tx0 - C2I_HALO);
+        if (ly >= 0 && ly < C2I_EDGE && lx >= 0 && lx < C2I_EDGE)
+          atomicAdd(&acc[ly * C2I_EDGE + lx], wgt[corner] * gval);
+        else  // large-offset outlier: device-scope fallback
+          atomicAdd(&gim[hh * g.W + ww], wgt[corner] * gval);
+      }
+    }
+  }
+  __syncthreads();
+  // cooperative flush: one global atomic per covered input pixel
+  for (int i = threadIdx.x; i < C2I_EDGE * C2I_EDGE; i += 256) {
+    const float v = acc[i];
+    if (v != 0.f) {
+      const int hh = ty0 - C2I_HALO + i / C2I_EDGE;
+      const int ww = tx0 - C2I_HALO + i % C2I_EDGE;
+      if (hh >= 0 && hh < g.H && ww >= 0 && ww < g.W)
+        atomicAdd(&gim[hh * g.W + ww], v);
+    }
+  }
+}
+
 // ------------------------------------------------------------- col2im
 // grad wrt input: distribute each column grad over its <=4 integer
 // neighbours with bilinear weights; atomicAdd into grad_im (device scope).
@@ -287,9 +373,18 @@ std::vector<at::Tensor> deform_conv2d_backward(
   // grad wrt columns: [B, C*K, HoWo] = W^T [C*K, Cout] x go2d (batched)
   auto col_grad = at::matmul(w2d.t(), go2d).contiguous();
 
-  // grad input (atomics)
+  // grad input (atomics; LDS-tiled variant for the 3x3/s1/p1 shape)
   auto grad_input = at::zeros_like(input);
-  {
+  if (g.kh == 3 && g.kw == 3 && g.sh == 1 && g.sw == 1 && g.ph == 1 &&
+      g.pw == 1 && g.dh == 1 && g.dw == 1) {
+    const int tiles = ((g.Ho + C2I_TILE - 1) / C2I_TILE) *
+                      ((g.Wo + C2I_TILE - 1) / C2I_TILE);
+    hipLaunchKernelGGL(dcn_col2im_tiled_kernel,
+                       dim3(tiles, g.C, g.B), dim3(256), 0, stream,
+                       col_grad.data_ptr<float>(), offset.data_ptr<float>(),
+                       mask.data_ptr<float>(), g,
+                       grad_input.data_ptr<float>());
+  } else {
     long long n = (long long)g.B * g.C * K * g.Ho * g.Wo;
     hipLaunchKernelGGL(dcn_col2im_kernel, dim3(esr_grid(n)), dim3(ESR_BLOCK),
                        0, stream, n, col_grad.data_ptr<float>(),
