@@ -373,10 +373,15 @@ std::vector<at::Tensor> deform_conv2d_backward(
   // grad wrt columns: [B, C*K, HoWo] = W^T [C*K, Cout] x go2d (batched)
   auto col_grad = at::matmul(w2d.t(), go2d).contiguous();
 
-  // grad input (atomics; LDS-tiled variant for the 3x3/s1/p1 shape)
+  // grad input (atomics; LDS-tiled variant for the 3x3/s1/p1 shape;
+  // ESR_DCN_TILED=0 forces the per-contribution kernel for A/B timing)
+  static const bool use_tiled = [] {
+    const char* e = getenv("ESR_DCN_TILED");
+    return e == nullptr || e[0] != '0';
+  }();
   auto grad_input = at::zeros_like(input);
-  if (g.kh == 3 && g.kw == 3 && g.sh == 1 && g.sw == 1 && g.ph == 1 &&
-      g.pw == 1 && g.dh == 1 && g.dw == 1) {
+  if (use_tiled && g.kh == 3 && g.kw == 3 && g.sh == 1 && g.sw == 1 &&
+      g.ph == 1 && g.pw == 1 && g.dh == 1 && g.dw == 1) {
     const int tiles = ((g.Ho + C2I_TILE - 1) / C2I_TILE) *
                       ((g.Wo + C2I_TILE - 1) / C2I_TILE);
     hipLaunchKernelGGL(dcn_col2im_tiled_kernel,
