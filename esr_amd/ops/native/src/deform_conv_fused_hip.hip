@@ -47,8 +47,9 @@ void dcn_fused_fwd_kernel(
   const int m0 = blockIdx.y * BM;
   const int b = blockIdx.z;
 
-  __shared__ float Bt[BK][BN + 1];   // im2col slice, +1 col pad
-  __shared__ float At[BM][BK + 1];   // weight slice
+  // double-buffered LDS: sample slice k0+16 while MFMA consumes slice k0
+  __shared__ float Bt[2][BK][BN + 1];   // im2col slices, +1 col pad
+  __shared__ float At[2][BM][BK + 1];   // weight slices
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -61,11 +62,8 @@ void dcn_fused_fwd_kernel(
   const float* off_b = offset + (long long)b * dg * 2 * KTAPS * HoWo;
   const float* msk_b = mask + (long long)b * dg * KTAPS * HoWo;
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    // ---- stage B-tile: each thread computes 4 im2col values ----
-    // thread t covers (krow = t/64, pixels t%64 + {0,64,128,192}/64...)
-    // layout: 4 k-rows per 64-pixel sweep -> t -> (kr = t>>6, px = t&63),
-    // iterate kr += 4.
+  auto stage = [&](int k0, int buf) {
+    // ---- stage B-tile: each thread samples 4 im2col values ----
     #pragma unroll
     for (int kr = wave; kr < BK; kr += 4) {
       const int k = k0 + kr;
@@ -96,33 +94,39 @@ void dcn_fused_fwd_kernel(
                  lh * (1 - lw) * v10 + lh * lw * v11) * m;
         }
       }
-      Bt[kr][lane] = val;
+      Bt[buf][kr][lane] = val;
     }
-    // ---- stage A-tile: weight[m0+row][k0+kc] ----
-    // 256 threads stage 64x16 = 1024 values: 4 per thread
+    // ---- stage A-tile: weight[m0+row][k0+kc] (4 values per thread) ----
     #pragma unroll
     for (int i = tid; i < BM * BK; i += 256) {
       const int row = i >> 4;         // /BK
       const int kc = i & 15;
       const int gm = m0 + row;
       const int gk = k0 + kc;
-      At[row][kc] = (gm < Cout && gk < K)
+      At[buf][row][kc] = (gm < Cout && gk < K)
           ? weight[(long long)gm * K + gk] : 0.f;
     }
-    __syncthreads();
+  };
 
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    if (k0 + BK < K)
+      stage(k0 + BK, cur ^ 1);        // overlaps the MFMAs below
     // ---- MFMA: 4 k-steps of 4, 4 pixel subtiles ----
     #pragma unroll
     for (int ks = 0; ks < 4; ++ks) {
       const int kk = ks * 4 + (lane >> 4);          // this lane's k
-      const float a = At[wave * 16 + (lane & 15)][kk];
+      const float a = At[cur][wave * 16 + (lane & 15)][kk];
       #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
-        const float bv = Bt[kk][nt * 16 + (lane & 15)];
+        const float bv = Bt[cur][kk][nt * 16 + (lane & 15)];
         acc[nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[nt], 0, 0, 0);
       }
     }
     __syncthreads();
+    cur ^= 1;
   }
 
   // ---- epilogue: D[i=row][j=col], col = lane&15, row = (lane>>4)*4 + r ----
