@@ -346,7 +346,18 @@ at::Tensor deform_conv2d_forward(
   TORCH_CHECK(mask.size(1) == g.dg * g.kh * g.kw, "mask channels");
   TORCH_CHECK(g.C % g.dg == 0, "C % deformable_groups != 0");
 
-  if (dcn_fused_applicable(input, weight, sh, sw, ph, pw, dh, dw, dg))
+  // Measured on gfx950 at the flagship shape (tools/bench_dcn.py,
+  // profiles/README.md): split im2col + hipBLASLt GEMM 0.46 ms vs fused
+  // im2col+MFMA 0.52 ms — the fused kernel's on-the-fly sampling is not
+  // hidden behind its MFMAs (hipcc drains the sampling loads at their
+  // immediate use).  Default to the faster split path; ESR_DCN_FUSED=1
+  // selects the fused kernel (numerics-equal, no column buffer in HBM).
+  static const bool use_fused = [] {
+    const char* e = getenv("ESR_DCN_FUSED");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (use_fused &&
+      dcn_fused_applicable(input, weight, sh, sw, ph, pw, dh, dw, dg))
     return deform_conv2d_forward_fused(input, offset, mask, weight, bias, dg);
 
   auto cols = dcn_im2col(input, offset, mask, g);
