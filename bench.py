@@ -99,6 +99,10 @@ def main():
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
     if world > 1:
+        # per-rank MIOpen find-db: 8 concurrent processes racing one user
+        # db file corrupts/serializes the find phase
+        os.environ.setdefault("MIOPEN_USER_DB_PATH",
+                              f"/tmp/miopen-rank{local_rank}")
         torch.cuda.set_device(local_rank)
         dist.init_process_group("nccl", init_method="env://")
     device = torch.device(f"cuda:{local_rank}")
@@ -226,22 +230,53 @@ def main():
         torch.cuda.synchronize()
         return graphs
 
+    def _rccl_capture_works() -> bool:
+        """Trial-capture ONE all-reduce and agree on the verdict across all
+        ranks, so every rank picks the same capture tier (a rank-divergent
+        tier would deadlock the collectives)."""
+        if world == 1:
+            return True
+        ok = 1
+        try:
+            probe = torch.ones(8, device=device)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                dist.all_reduce(probe)
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                dist.all_reduce(probe)
+            torch.cuda.synchronize()
+            g.replay()
+            torch.cuda.synchronize()
+        except Exception:
+            ok = 0
+        verdict = torch.tensor([ok], device=device)
+        dist.all_reduce(verdict, op=dist.ReduceOp.MIN)
+        return bool(verdict.item())
+
     # Tiered capture: (a) whole step incl. RCCL all-reduce + Adam in the
     # graph; (b) fwd+bwd in the graph, comm + Adam eager; (c) fully eager.
     graphs = None
     graph_mode = "eager"
     if not args.no_graphs:
+        comm_in_graph = _rccl_capture_works()
+
         def full_step():
             fwd_bwd()
             comm()
             opt_step()
         try:
+            if not comm_in_graph:
+                raise RuntimeError("RCCL graph capture unavailable")
             _warmup_side_stream(full_step)
             graphs = _capture(full_step)
             graph_mode = "full"
         except Exception as e:
-            print(f"[bench] full-step capture failed ({type(e).__name__}); "
-                  f"trying fwd+bwd-only capture", flush=True)
+            print(f"[bench] full-step capture unavailable ({type(e).__name__});"
+                  f" trying fwd+bwd-only capture", flush=True)
             torch.cuda.synchronize()
             try:
                 _warmup_side_stream(fwd_bwd)
