@@ -354,15 +354,29 @@ class ESRNet(nn.Module):
         slice — no gather, no copy.
         """
         B, L, C, H, W = frames.shape
+        nW = L - seqn + 1
         x, box = _pad_multiple(frames, self.DOWN_SCALE)
         x = x.transpose(0, 1).reshape(L * B, C, x.size(-2), x.size(-1))
         x = self.head(x)
         feats_list = self.feat_extract(x)
-        outs = []
-        for w in range(L - seqn + 1):
-            sl = slice(w * B, (w + seqn) * B)
-            deep = self.time_propagate.forward_frames(feats_list[0][sl], seqn)
-            out = self.spacetime_fuse.forward_frames(
-                deep, [f[sl] for f in feats_list], seqn)
-            outs.append(_crop(self.tail(out), box))
-        return outs
+
+        # temporal propagation is stateful -> sequential over windows
+        deeps = [self.time_propagate.forward_frames(
+            feats_list[0][w * B:(w + seqn) * B], seqn) for w in range(nW)]
+
+        # spatio-temporal fusion + decoder have NO cross-window state:
+        # run ALL windows in one batched pass (combined batch nW*B,
+        # frame-major: frame n block = that frame of every window)
+        def window_major(per_window, n):
+            return torch.cat([d[n * B:(n + 1) * B] for d in per_window], dim=0)
+
+        deep_all = torch.cat([window_major(deeps, n) for n in range(seqn)],
+                             dim=0)
+        feats_all = [
+            torch.cat([torch.cat([f[(w + n) * B:(w + n + 1) * B]
+                                  for w in range(nW)], dim=0)
+                       for n in range(seqn)], dim=0)
+            for f in feats_list]
+        out_all = self.spacetime_fuse.forward_frames(deep_all, feats_all, seqn)
+        out_all = _crop(self.tail(out_all), box)
+        return [out_all[w * B:(w + 1) * B] for w in range(nW)]
