@@ -150,3 +150,13 @@ def test_epoch_based_training(tmp_path, synth_datalist):
     assert ckpts, "no epoch checkpoint written"
     state = torch.load(ckpts[-1], map_location="cpu", weights_only=False)
     assert state["trainer"]["training_mode"] == "epoch_based_train"
+
+
+def test_resumer_rejects_wrong_component_name(trained, tmp_path):
+    from esr_amd.engine.checkpoint import Resumer
+    parser, trainer = trained
+    ckpt = sorted(Path(parser.save_dir).glob("checkpoint-iteration*.pth"))[-1]
+    bad_cfg = {"model": {"name": "SomethingElse"}}
+    r = Resumer(str(ckpt), config=bad_cfg)
+    with pytest.raises(ValueError):
+        r.resume_model(trainer.model)

@@ -113,3 +113,39 @@ def test_forward_sequence_matches_per_window():
     assert abs(loss_seq.item() - loss_ref.item()) < 1e-6
     for ga, gb in zip(grads_seq, grads_ref):
         assert torch.allclose(ga, gb, atol=1e-5), (ga - gb).abs().max().item()
+
+
+def test_gtc_frozen_stateless():
+    torch.manual_seed(4)
+    m = build_model("ESRNet", inch=2, basech=8, num_frame=3, gtc_frozen=True)
+    x = torch.rand(1, 3, 2, 16, 16)
+    m.reset_states()
+    y1 = m(x)
+    assert m.time_propagate.state is None     # frozen: never persists
+    y2 = m(x)
+    assert torch.allclose(y1, y2, atol=1e-6)  # stateless => identical
+
+
+def test_convlstm_recurrent_type():
+    torch.manual_seed(5)
+    m = build_model("ESRNet", inch=2, basech=8, num_frame=3,
+                    recurrent_block_type="convlstm")
+    y = m(torch.rand(1, 3, 2, 16, 16))
+    assert y.shape == (1, 2, 16, 16)
+    state = m.time_propagate.state
+    assert isinstance(state, tuple) and len(state) == 2  # (h, c)
+    m(torch.rand(1, 3, 2, 16, 16))
+    m.detach_states()
+    assert not m.time_propagate.state[0].requires_grad
+
+
+def test_detach_states_truncates_bptt():
+    torch.manual_seed(6)
+    m = build_model("ESRNet", inch=2, basech=8, num_frame=3)
+    m.reset_states()
+    m(torch.rand(1, 3, 2, 16, 16))
+    m.detach_states()
+    loss = (m(torch.rand(1, 3, 2, 16, 16)) ** 2).mean()
+    loss.backward()   # must not error about freed graphs from window 1
+    assert all(torch.isfinite(p.grad).all()
+               for p in m.parameters() if p.grad is not None)
