@@ -11,7 +11,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 __all__ = ["SelfAttention", "ConvLayer1D", "ConvLayer3D", "InceptionBlock",
-           "DilatedBlock", "knn", "DenseEdgeConv"]
+           "DilatedBlock", "knn", "DenseEdgeConv", "FeedbackBlock"]
 
 
 class SelfAttention(nn.Module):
@@ -151,3 +151,49 @@ class DenseEdgeConv(nn.Module):
             y = F.relu(conv(x))
             x = torch.cat([x, y], dim=1)
         return x.max(dim=-1).values.transpose(1, 2)              # [B,N,C_out]
+
+
+class FeedbackBlock(nn.Module):
+    """SRFBN-style feedback block: iterative up/down projection pairs with
+    dense connections (parity: ESR:models/submodules.py:755-871, the SRFBN
+    helper family used as an SR baseline)."""
+
+    def __init__(self, channels, num_groups=3, scale=2):
+        super().__init__()
+        k, s, p = {2: (6, 2, 2), 4: (8, 4, 2), 8: (12, 8, 2)}[scale]
+        self.compress_in = nn.Conv2d(2 * channels, channels, 1)
+        self.up_blocks = nn.ModuleList()
+        self.down_blocks = nn.ModuleList()
+        self.uptran = nn.ModuleList()
+        self.downtran = nn.ModuleList()
+        for i in range(num_groups):
+            self.up_blocks.append(
+                nn.ConvTranspose2d(channels, channels, k, s, p))
+            self.down_blocks.append(nn.Conv2d(channels, channels, k, s, p))
+            if i > 0:
+                self.uptran.append(nn.Conv2d((i + 1) * channels, channels, 1))
+                self.downtran.append(nn.Conv2d((i + 1) * channels, channels, 1))
+        self.compress_out = nn.Conv2d(num_groups * channels, channels, 1)
+        self.last_hidden = None
+
+    def reset_state(self):
+        self.last_hidden = None
+
+    def forward(self, x):
+        if self.last_hidden is None:
+            self.last_hidden = torch.zeros_like(x)
+        x = self.compress_in(torch.cat([x, self.last_hidden], dim=1))
+        lows, highs = [x], []
+        for i, (up, down) in enumerate(zip(self.up_blocks, self.down_blocks)):
+            lo = torch.cat(lows, dim=1)
+            if i > 0:
+                lo = self.uptran[i - 1](lo)
+            hi = F.relu(up(lo))
+            highs.append(hi)
+            hi_cat = torch.cat(highs, dim=1)
+            if i > 0:
+                hi_cat = self.downtran[i - 1](hi_cat)
+            lows.append(F.relu(down(hi_cat)))
+        out = self.compress_out(torch.cat(lows[1:], dim=1))
+        self.last_hidden = out
+        return out

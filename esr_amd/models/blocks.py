@@ -244,3 +244,19 @@ class MLP(nn.Module):
         for i, layer in enumerate(self.layers):
             x = F.relu(layer(x)) if i < self.num_layers - 1 else layer(x)
         return x
+
+
+def recursive_clone(tensor_or_state):
+    """Deep-clone nested tensors/tuples/lists (parity:
+    ESR:models/model_util.py:208-229 copy_states/recursive_clone)."""
+    if isinstance(tensor_or_state, torch.Tensor):
+        return tensor_or_state.clone()
+    if isinstance(tensor_or_state, (tuple, list)):
+        return type(tensor_or_state)(recursive_clone(t)
+                                     for t in tensor_or_state)
+    return tensor_or_state
+
+
+def copy_states(states):
+    """Snapshot recurrent states for later restoration."""
+    return recursive_clone(states)

@@ -40,3 +40,17 @@ def test_dense_edge_conv():
     y = m(torch.rand(2, 32, 8))
     assert y.shape == (2, 32, 2 * 8 + 3 * 12)
     y.mean().backward()
+
+
+def test_feedback_block_recurrence():
+    import torch
+    from esr_amd.models.aux_blocks import FeedbackBlock
+    fb = FeedbackBlock(8, num_groups=3, scale=2)
+    x = torch.rand(1, 8, 16, 16)
+    y1 = fb(x)
+    y2 = fb(x)  # hidden feedback -> different output on same input
+    assert y1.shape == (1, 8, 16, 16)
+    assert not torch.allclose(y1, y2)
+    fb.reset_state()
+    y3 = fb(x)
+    assert torch.allclose(y1, y3, atol=1e-6)
