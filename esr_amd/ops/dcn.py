@@ -100,9 +100,16 @@ class _DeformConvHIP(torch.autograd.Function):
     def forward(ctx, input, offset, mask, weight, bias,
                 stride, padding, dilation, deformable_groups):
         ext = get_ext()
+        # save the CONTIGUOUS tensors: the HIP backward indexes its inputs
+        # with contiguous strides (DeformAlign2d's offset is a channel slice
+        # and non-contiguous for B>1 — saving the original silently corrupts
+        # the backward)
+        input = input.contiguous()
+        offset = offset.contiguous()
+        mask = mask.contiguous()
+        weight = weight.contiguous()
         out = ext.deform_conv2d_forward(
-            input.contiguous(), offset.contiguous(), mask.contiguous(),
-            weight.contiguous(), bias,
+            input, offset, mask, weight, bias,
             stride[0], stride[1], padding[0], padding[1],
             dilation[0], dilation[1], deformable_groups)
         ctx.save_for_backward(input, offset, mask, weight)

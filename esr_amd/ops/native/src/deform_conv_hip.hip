@@ -375,6 +375,10 @@ std::vector<at::Tensor> deform_conv2d_backward(
     const at::Tensor& weight, const at::Tensor& grad_out,
     int64_t sh, int64_t sw, int64_t ph, int64_t pw, int64_t dh, int64_t dw,
     int64_t dg) {
+  TORCH_CHECK(input.is_contiguous() && offset.is_contiguous() &&
+              mask.is_contiguous() && weight.is_contiguous() &&
+              grad_out.is_contiguous(),
+              "deform_conv2d_backward: contiguous tensors required");
   auto g = make_geom(input, weight, sh, sw, ph, pw, dh, dw, dg);
   const int K = g.kh * g.kw;
   auto stream = at::hip::getCurrentHIPStream();

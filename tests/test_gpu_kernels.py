@@ -304,3 +304,38 @@ class TestFusedDCN:
         hook = ext.deform_conv2d_forward_fused(input, offset, mask, weight,
                                                bias, 8)
         assert torch.allclose(pub, hook, atol=1e-4, rtol=1e-4)
+
+
+class TestDeformAlignGrads:
+    def test_module_grads_match_cpu_oracle_batch2(self, ext):
+        """DeformAlign2d backward vs the CPU autograd oracle, batch 2: the
+        offset channel-slice is NON-contiguous for B>1 — this is the
+        regression test for the saved-tensor contiguity bug."""
+        from esr_amd.ops.dcn import DeformAlign2d
+        torch.manual_seed(0)
+        m = DeformAlign2d(8, 8, 3, stride=1, padding=1, deformable_groups=2)
+        # non-zero offsets so the deformable path is exercised
+        with torch.no_grad():
+            m.conv_offset_mask.weight.normal_(0, 0.1)
+            m.conv_offset_mask.bias.normal_(0, 0.5)
+        x = torch.randn(2, 8, 12, 12)
+        feat = torch.randn(2, 8, 12, 12)
+
+        m_cpu = m
+        out = m_cpu(x, feat)
+        loss = out.square().mean()
+        loss.backward()
+        grads_cpu = {n: p.grad.clone() for n, p in m_cpu.named_parameters()}
+        for p in m_cpu.parameters():
+            p.grad = None
+
+        m_gpu = m.cuda()
+        out_gpu = m_gpu(x.cuda(), feat.cuda())
+        assert torch.allclose(out_gpu.cpu(), out, atol=1e-4)
+        loss_gpu = out_gpu.square().mean()
+        loss_gpu.backward()
+        for n, p in m_gpu.named_parameters():
+            ref = grads_cpu[n]
+            err = (p.grad.cpu() - ref).abs().max().item()
+            scale = ref.abs().max().item() + 1e-8
+            assert err / scale < 2e-3, f"{n}: rel err {err/scale:.3e}"
