@@ -139,3 +139,28 @@ def test_store_events_dtype_roundtrip(tmp_path):
     s = EventStore(tmp_path / "t.evs")
     ev = s.events("ori", 0, 2)
     assert ev[0, 1] == 65535 and ev[3, 1] == -1 and ev[2, 1] == 1.5
+
+
+def test_ssim_psnr_sanity():
+    from esr_amd.loss import ssim, psnr, rmse, mse
+    g = torch.Generator().manual_seed(1)
+    a = torch.rand(2, 16, 16, generator=g)
+    assert abs(ssim(a, a) - 1.0) < 1e-6           # identical -> 1
+    noisy = a + 0.2 * torch.randn(2, 16, 16, generator=g)
+    noisier = a + 0.6 * torch.randn(2, 16, 16, generator=g)
+    assert ssim(a, noisy) > ssim(a, noisier)      # monotone in noise
+    assert psnr(noisy, a) > psnr(noisier, a)
+    assert abs(rmse(noisy, a) ** 2 - mse(noisy, a)) < 1e-5
+    assert psnr(a, a) == float("inf")
+
+
+def test_lpips_deterministic_and_discriminative():
+    from esr_amd.loss import PerceptualLoss
+    g = torch.Generator().manual_seed(2)
+    a = torch.rand(1, 2, 32, 32, generator=g)
+    b = torch.rand(1, 2, 32, 32, generator=g)
+    p1 = PerceptualLoss(net="alex")
+    p2 = PerceptualLoss(net="alex")          # same seed -> same projection
+    assert abs(p1(a, b).item() - p2(a, b).item()) < 1e-6
+    assert p1(a, a).item() < 1e-6
+    assert p1(a, b).item() > p1(a, a).item()
