@@ -82,7 +82,7 @@ def _ssim_single(p: torch.Tensor, t: torch.Tensor, data_range: float,
     C1 = (K1 * data_range) ** 2
     C2 = (K2 * data_range) ** 2
     num = (2 * ux * uy + C1) * (2 * vxy + C2)
-    den = (ux * ux + uy * uy + C1) * (vx * vy + C2)
+    den = (ux * ux + uy * uy + C1) * (vx + vy + C2)
     return (num / den).mean().item()
 
 
