@@ -150,3 +150,49 @@ def test_scaled_encoding_resplat():
     up = scaled_count_encoding(norm, (8, 8), "cnt")
     assert up.shape == (2, 8, 8)
     assert up[0, 2, 2] == 1 and up[1, 6, 6] == 1
+
+
+def test_voxel_matches_loop_oracle():
+    """Temporal-bilinear voxel vs a per-event loop (independent oracle)."""
+    g = torch.Generator().manual_seed(11)
+    n, B, H, W = 200, 5, 6, 7
+    xs = (torch.rand(n, generator=g) * W).floor()
+    ys = (torch.rand(n, generator=g) * H).floor()
+    ts = torch.rand(n, generator=g)
+    ps = torch.randint(0, 2, (n,), generator=g).float() * 2 - 1
+    v = events_to_voxel(xs, ys, ts, ps, B, (H, W))
+    ref = torch.zeros(B, H, W)
+    for x, y, t, p in zip(xs.tolist(), ys.tolist(), ts.tolist(), ps.tolist()):
+        for b in range(B):
+            w = max(0.0, 1.0 - abs(t * (B - 1) - b))
+            ref[b, int(y), int(x)] += p * w
+    assert torch.allclose(v, ref, atol=1e-4), (v - ref).abs().max().item()
+
+
+def test_redistribute_polarity_stack_5d():
+    g = torch.Generator().manual_seed(12)
+    stack = torch.randint(0, 4, (2, 2, 3, 4, 4), generator=g).float()
+    stack[:, 1] *= -1  # negative polarity plane holds negative values
+    cloud = redistribute_stack(stack, mode="linear")
+    for b in range(2):
+        ev = cloud[b]
+        ev = ev[ev.abs().sum(1) > 0]
+        assert ev.shape[0] == int(stack[b].abs().sum().item())
+        # channel-0 cells (positive values) emit +1, channel-1 emit -1
+        n_pos = int(stack[b, 0].abs().sum().item())
+        assert (ev[:, 3] > 0).sum().item() == n_pos
+
+
+def test_scaled_encoding_stack_and_events_modes():
+    # >3 events (the reference's degenerate early-out returns zeros below)
+    ev = torch.tensor([[1.0, 1.0, 3.0, 3.0, 2.0],
+                       [1.0, 1.0, 3.0, 3.0, 2.0],
+                       [0.1, 0.2, 0.8, 0.9, 0.5],
+                       [1.0, 1.0, -1.0, -1.0, 1.0]])
+    norm = normalize_events(ev, (4, 4))
+    st = scaled_count_encoding(norm, (8, 8), "stack", time_bins=2)
+    assert st.shape == (2, 8, 8)
+    assert st[0, 2, 2] == 2 and st[1, 6, 6] == -2
+    evm = scaled_count_encoding(norm, (8, 8), "events")
+    assert evm.shape == (4, 5)
+    assert evm[0, 0] == 2 and evm[1, 2] == 6

@@ -99,3 +99,15 @@ def test_sobel_shapes():
     # constant image -> zero gradients
     gx, gy = s(torch.ones(1, 1, 8, 8))
     assert gx.abs().max() < 1e-6 and gy.abs().max() < 1e-6
+
+
+def test_deblur_events_bilinear_path():
+    """round_idx=False (bilinear) path incl. the 4x polarity-mask repeat."""
+    ev, pol = _events(B=1, N=32)
+    flow = torch.rand(1, 2, 16, 16) * 0.005
+    iwe = deblur_events(flow, ev, (16, 16), flow_scaling=16,
+                        round_idx=False, polarity_mask=pol[:, :, 0:1])
+    assert iwe.shape == (1, 1, 16, 16)
+    # bilinear weights of in-range positive events sum to <= n_pos
+    n_pos = pol[:, :, 0].sum().item()
+    assert 0 < iwe.sum().item() <= n_pos + 1e-4
