@@ -181,3 +181,46 @@ def test_sequence_pause_chain(store_path):
     for item in seq[1:]:
         assert item["inp_cnt"].sum() == 0
         assert item["gt_cnt"].sum() > 0   # GT stays (frozen index)
+
+
+def test_frame_mode_windowing(store_path):
+    cfg = _ds_config(mode="frame")
+    cfg["need_gt_frame"] = True
+    ds = EventSRDataset(store_path, cfg)
+    assert len(ds) == ds.store.num_images - 1
+    item = ds[0]
+    assert item["frame"].shape[0] == 1          # the mode's aligned frame
+    assert item["inp_cnt"].sum() > 0
+
+
+def test_custom_resolution_outputs(store_path):
+    cfg = _ds_config(custom_resolution=[24, 24])
+    ds = EventSRDataset(store_path, cfg)
+    item = ds[0]
+    assert item["inp_custom_cnt"].shape == (2, 24, 24)
+    assert item["inp_custom_scaled_cnt"].shape == (2, 48, 48)
+    assert item["inp_custom_down_cnt"].shape == (2, 12, 12)
+    assert item["gt_custom_cnt"].shape == (2, 48, 48)
+    # rounded bicubic values are integers
+    assert torch.allclose(item["inp_custom_cnt"],
+                          item["inp_custom_cnt"].round())
+
+
+def test_flat_event_loader(synth_datalist):
+    from esr_amd.data import make_event_loader
+    cfg = {
+        "use_ddp": False, "path_to_datalist_txt": synth_datalist,
+        "batch_size": 3, "shuffle": False, "num_workers": 0,
+        "pin_memory": False, "drop_last": True,
+        "dataset": _ds_config(window=1024, sliding_window=512),
+    }
+    loader = make_event_loader(cfg)
+    batch = next(iter(loader))
+    assert batch["inp_scaled_cnt"].shape[0] == 3
+    assert batch["inp_scaled_cnt"].dim() == 4
+
+
+def test_redistribute_all_empty():
+    from esr_amd.ops import redistribute_stack
+    out = redistribute_stack(torch.zeros(3, 4, 5, 5))
+    assert out.shape == (3, 1, 4) and out.abs().sum() == 0
