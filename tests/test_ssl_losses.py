@@ -111,3 +111,26 @@ def test_deblur_events_bilinear_path():
     # bilinear weights of in-range positive events sum to <= n_pos
     n_pos = pol[:, :, 0].sum().item()
     assert 0 < iwe.sum().item() <= n_pos + 1e-4
+
+
+def test_averaged_iwe_divides_by_distinct_sources():
+    """Two source pixels warped onto one destination: the averaged IWE is
+    raw_count / n_distinct_sources (parity: ESR:loss/flow.py:113-232)."""
+    H = W = 8
+    scaling = max(H, W)
+    # events at (y=0,x=0) and (y=0,x=2), both positive, ts=0
+    ev = torch.tensor([[[0.0, 0.0, 0.0, 1.0],
+                        [0.0, 0.0, 2.0, 1.0]]])   # (ts, y, x, p)
+    pol = torch.tensor([[[1.0, 0.0], [1.0, 0.0]]])
+    flow = torch.zeros(1, 2, H, W)
+    flow[0, 0, 0, 0] = 1.0 / scaling    # x-flow at (0,0): +1 px
+    flow[0, 0, 0, 2] = -1.0 / scaling   # x-flow at (0,2): -1 px
+    avg = AveragedIWE((H, W))
+    out = avg(flow, ev, pol)
+    assert out[0, 0, 0, 1].item() == 1.0      # 2 events / 2 sources
+    assert out[0, 0].sum().item() == 1.0
+    # same flow for both events from ONE source pixel -> no averaging
+    ev2 = torch.tensor([[[0.0, 0.0, 0.0, 1.0],
+                         [0.0, 0.0, 0.0, 1.0]]])
+    out2 = avg(flow, ev2, pol)
+    assert out2[0, 0, 0, 1].item() == 2.0     # 2 events / 1 source
