@@ -98,6 +98,9 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", 1))
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    assert world == args.gpus, (
+        f"--gpus {args.gpus} but WORLD_SIZE={world}: launch N>1 via torchrun "
+        f"(python -m torch.distributed.run --nproc-per-node N bench.py --gpus N)")
     if world > 1:
         # per-rank MIOpen find-db: 8 concurrent processes racing one user
         # db file corrupts/serializes the find phase

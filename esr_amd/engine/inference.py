@@ -30,14 +30,16 @@ _IMG_DIRS = ["lr_event_img", "hr_scaled_event_img", "hr_esr_event_img",
              "hr_bicubic_event_img", "hr_gt_event_img"]
 
 
-def build_metrics(device="cpu", lpips_net="alex", lpips_weights=None):
+def build_metrics(device="cpu", lpips_net="alex", lpips_weights=None,
+                  lpips_backbone=None):
     return {
         "l1": l1_fn,
         "mse": mse_fn,
         "ssim": ssim_fn,
         "psnr": psnr_fn,
         "lpips": PerceptualLoss(net=lpips_net, device=device,
-                                weights_path=lpips_weights),
+                                weights_path=lpips_weights,
+                                backbone_path=lpips_backbone),
     }
 
 
@@ -126,6 +128,14 @@ def infer_sequence(dataloader_config, data_path, model, device,
                            True, str(img_path / "gt_img" / f"{i:09d}.png"))
 
     result = track.result()
+    lp = metrics.get("lpips")
+    if lp is not None and hasattr(lp, "paper_comparable") \
+            and not lp.paper_comparable:
+        # trained linear heads ship in-repo, but without an ImageNet backbone
+        # checkpoint the feature extractor is a seeded random projection —
+        # flag it so lpips numbers are not compared against paper tables
+        result["lpips_note"] = ("random-init backbone (no ImageNet checkpoint"
+                                " in env): lpips values are relative-only")
     if output_path is not None:
         with open(Path(output_path) / "results.yml", "w") as f:
             yaml.safe_dump({"evaluation results": result}, f)

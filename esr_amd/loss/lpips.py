@@ -5,19 +5,32 @@ backbones with bundled linear-head weights
 (ESR:loss/PerceptualSimilarity/models/networks_basic.py:32-101,
 ESR:loss/restore.py:10-39).  This environment has neither torchvision nor
 network access, so the backbones (AlexNet / VGG16 feature stacks) are
-implemented here directly.  Weights are random-initialized unless a
-state-dict path is supplied — the metric is then a fixed random-projection
-perceptual distance, which is deterministic (seeded) and usable for
-relative comparisons in tests/CI; load real weights for paper-comparable
-LPIPS numbers.
+implemented here directly.
+
+Weight status:
+  * the LEARNED LINEAR HEADS ship with this repo (esr_amd/loss/weights/
+    {alex,vgg}.pth — the reference's v0.1 head weights,
+    ESR:loss/PerceptualSimilarity/models/weights/v0.1) and load by default;
+  * the BACKBONE conv weights still need an ImageNet checkpoint
+    (torchvision layout, pass via ``backbone_path``).  Without one the
+    backbone is a fixed seeded random projection: deterministic and usable
+    for relative comparisons, but NOT paper-comparable — callers can check
+    ``LPIPS.backbone_pretrained`` and annotate their outputs.
 """
 
 from __future__ import annotations
+
+from pathlib import Path
 
 import torch
 import torch.nn as nn
 
 __all__ = ["PerceptualLoss", "LPIPS"]
+
+_BUNDLED_HEADS = {
+    "alex": Path(__file__).resolve().parent / "weights" / "alex.pth",
+    "vgg": Path(__file__).resolve().parent / "weights" / "vgg.pth",
+}
 
 
 class _AlexFeatures(nn.Module):
@@ -75,6 +88,26 @@ class _VGG16Features(nn.Module):
         return outs
 
 
+def _map_torchvision_backbone(net, sd):
+    """Map a torchvision ``alexnet``/``vgg16`` ``features.N.*`` state dict
+    onto the slice layout above (conv layers zipped in order)."""
+    idxs = sorted({int(k.split(".")[1]) for k in sd
+                   if k.startswith("features.") and k.endswith(".weight")})
+    convs = [(sd[f"features.{i}.weight"], sd.get(f"features.{i}.bias"))
+             for i in idxs]
+    out, n = {}, 0
+    proto = {"alex": _AlexFeatures, "vgg": _VGG16Features}[net]()
+    for name, mod in proto.named_modules():
+        if isinstance(mod, nn.Conv2d):
+            w, b = convs[n]
+            out[f"{name}.weight"] = w
+            if b is not None:
+                out[f"{name}.bias"] = b
+            n += 1
+    assert n == len(convs), f"backbone conv count mismatch: {n} vs {len(convs)}"
+    return out
+
+
 def _normalize_tensor(x, eps=1e-10):
     norm = torch.sqrt(torch.sum(x ** 2, dim=1, keepdim=True))
     return x / (norm + eps)
@@ -90,22 +123,64 @@ class LPIPS(nn.Module):
     SCALE = [0.458, 0.448, 0.450]
 
     def __init__(self, net: str = "alex", weights_path: str | None = None,
-                 seed: int = 1234):
+                 seed: int = 1234, backbone_path: str | None = None):
         super().__init__()
-        torch.manual_seed(seed)  # deterministic random projection fallback
-        self.features = {"alex": _AlexFeatures, "vgg": _VGG16Features}[net]()
-        self.lins = nn.ModuleList([
-            nn.Conv2d(c, 1, 1, bias=False) for c in self.features.CH])
-        for lin in self.lins:
-            nn.init.uniform_(lin.weight, 0.0, 0.1)  # nonneg like trained heads
+        # local generator + fork_rng: constructing a metric must neither
+        # reseed nor advance the global torch RNG (advisor finding r1);
+        # nn.Conv2d default init would otherwise consume global draws
+        gen = torch.Generator().manual_seed(seed)
+        with torch.random.fork_rng(devices=[]):
+            self.features = {"alex": _AlexFeatures,
+                             "vgg": _VGG16Features}[net]()
+            self.lins = nn.ModuleList([
+                nn.Conv2d(c, 1, 1, bias=False) for c in self.features.CH])
+        self.backbone_pretrained = False
+        with torch.no_grad():
+            for p in self.features.parameters():
+                if p.dim() > 1:
+                    # kaiming-uniform equivalent drawn from the local gen
+                    fan_in = p[0].numel()
+                    bound = (6.0 / fan_in) ** 0.5
+                    p.uniform_(-bound, bound, generator=gen)
+                else:
+                    p.zero_()
+        if backbone_path:
+            self.features.load_state_dict(
+                _map_torchvision_backbone(
+                    net, torch.load(backbone_path, map_location="cpu")))
+            self.backbone_pretrained = True
+        head_sd = None
         if weights_path:
-            sd = torch.load(weights_path, map_location="cpu")
-            self.load_state_dict(sd, strict=False)
+            head_sd = torch.load(weights_path, map_location="cpu")
+        elif _BUNDLED_HEADS.get(net, Path("/nonexistent")).exists():
+            # the reference's trained v0.1 linear heads, bundled as data
+            head_sd = torch.load(str(_BUNDLED_HEADS[net]), map_location="cpu")
+        if head_sd is not None:
+            self._load_heads(head_sd)
+        else:
+            for lin in self.lins:
+                with torch.no_grad():
+                    lin.weight.uniform_(0.0, 0.1, generator=gen)
+        self.heads_pretrained = head_sd is not None
         self.register_buffer("shift", torch.tensor(self.SHIFT).view(1, 3, 1, 1))
         self.register_buffer("scale", torch.tensor(self.SCALE).view(1, 3, 1, 1))
         for p in self.parameters():
             p.requires_grad_(False)
         self.eval()
+
+    def _load_heads(self, sd):
+        """Load linear-head weights from either this module's own layout
+        (``lins.N.weight``) or the reference's
+        (``linN.model.1.weight`` — ESR:loss/PerceptualSimilarity/models/
+        networks_basic.py:32-101)."""
+        with torch.no_grad():
+            for i, lin in enumerate(self.lins):
+                for key in (f"lins.{i}.weight", f"lin{i}.model.1.weight"):
+                    if key in sd:
+                        lin.weight.copy_(sd[key])
+                        break
+                else:
+                    raise KeyError(f"no head weight for slice {i} in state dict")
 
     def forward(self, pred, target, normalize=True):
         if normalize:  # [0,1] -> [-1,1]
@@ -126,9 +201,16 @@ class PerceptualLoss:
     """N-channel wrapper (parity: ESR:loss/restore.py:10-39): 1-ch inputs are
     replicated to RGB; >3-ch inputs are scored per channel and averaged."""
 
-    def __init__(self, weight=1.0, net="alex", device="cpu", weights_path=None):
-        self.model = LPIPS(net=net, weights_path=weights_path).to(device)
+    def __init__(self, weight=1.0, net="alex", device="cpu", weights_path=None,
+                 backbone_path=None):
+        self.model = LPIPS(net=net, weights_path=weights_path,
+                           backbone_path=backbone_path).to(device)
         self.weight = weight
+
+    @property
+    def paper_comparable(self):
+        """True only when both backbone and heads carry trained weights."""
+        return self.model.backbone_pretrained and self.model.heads_pretrained
 
     @torch.no_grad()
     def __call__(self, pred, target, normalize=True):

@@ -28,7 +28,7 @@ __global__ void splat_count_kernel(long long n, int N, int H, int W,
     const int yi = (int)e.y;
     if (e.w == 0.f || xi < 0 || xi >= W || yi < 0 || yi >= H) continue;
     const int ch = e.w > 0.f ? 0 : 1;
-    const float wgt = e.w > 0.f ? e.w * e.w : e.w * e.w;  // ps*relu / ps*min
+    const float wgt = fabsf(e.w);  // count magnitude (p = +/-1 -> 1 per event)
     atomicAdd(&out[(((long long)b * 2 + ch) * H + yi) * W + xi], wgt);
   }
 }
