@@ -212,12 +212,25 @@ def main():
         static_gts[0] = list(gts)
 
     def _warmup_side_stream(body):
+        """Warmup initializes MIOpen plans + capturable-Adam state tensors,
+        but its real optimizer.step()s run on garbage data — snapshot the
+        weights first and roll back (in place) before capture."""
+        snap_params = params + (masters if pure_bf16 else [])
+        snap = [prm.detach().clone() for prm in snap_params]
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(3):
                 body()
         torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        with torch.no_grad():
+            for prm, sv in zip(snap_params, snap):
+                prm.copy_(sv)
+            for state in optimizer.state.values():
+                for v in state.values():
+                    if torch.is_tensor(v):
+                        v.zero_()
         torch.cuda.synchronize()
 
     def _capture(body):

@@ -95,13 +95,37 @@ class GraphedBPTTStep:
         self.optimizer.step()
         return loss, mse
 
+    def _snapshot_state(self):
+        params = [p for g in self.optimizer.param_groups for p in g["params"]]
+        return [p.detach().clone() for p in params], params
+
+    def _restore_state(self, snapshot, params):
+        """Undo the warmup steps: params back to their pre-warmup values and
+        the optimizer state zeroed in place (= never stepped).  In-place so
+        the state tensors the capture records keep their addresses."""
+        with torch.no_grad():
+            for p, s in zip(params, snapshot):
+                p.copy_(s)
+            for state in self.optimizer.state.values():
+                for v in state.values():
+                    if torch.is_tensor(v):
+                        v.zero_()
+
     def capture(self):
+        # warmup on a side stream initializes cuDNN/MIOpen plans, autograd
+        # graph allocations and the capturable optimizer's state TENSORS —
+        # but the real optimizer.step()s it runs would perturb the model
+        # with garbage-gradient updates (advisor finding r1), so snapshot
+        # first and roll back before recording the graph.
+        snapshot, params = self._snapshot_state()
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(self._warmup):
                 self._body()
         torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        self._restore_state(snapshot, params)
         torch.cuda.synchronize()
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):

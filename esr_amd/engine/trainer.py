@@ -202,8 +202,29 @@ class Trainer:
                 loss = loss + mse_loss
         if train:
             loss.backward()
+            self._sync_grads_if_needed()
             self.optimizer.step()
         return loss.detach(), mse_loss.detach(), pred.detach()
+
+    def _sync_grads_if_needed(self):
+        """Manual gradient all-reduce for the eager path when the model is
+        not DDP-wrapped (hip_graphs mode skips wrap_ddp; if graph capture
+        fails at runtime the eager fallback must still synchronize ranks —
+        advisor finding r1)."""
+        from ..parallel import get_world_size
+        world = get_world_size()
+        if world <= 1 or hasattr(self.model, "module"):
+            return
+        import torch.distributed as dist
+        grads = [p.grad for p in self.model.parameters()
+                 if p.requires_grad and p.grad is not None]
+        if not grads:
+            return
+        flat = torch._utils._flatten_dense_tensors(grads)
+        dist.all_reduce(flat)
+        flat.div_(world)
+        for g, s in zip(grads, torch._utils._unflatten_dense_tensors(flat, grads)):
+            g.copy_(s)
 
     # ---------------- loops ----------------
 

@@ -99,10 +99,13 @@ def main():
 
         mean = {}
         if all_results:
-            keys = next(iter(all_results.values())).keys()
-            for k in keys:
-                mean[k] = sum(r[k] for r in all_results.values()) \
-                    / len(all_results)
+            first = next(iter(all_results.values()))
+            for k, v in first.items():
+                if isinstance(v, str):   # annotations (e.g. lpips_note)
+                    mean[k] = v
+                else:
+                    mean[k] = sum(r[k] for r in all_results.values()) \
+                        / len(all_results)
         mdir.mkdir(parents=True, exist_ok=True)
         with open(mdir / "mean_results.yml", "w") as f:
             yaml.safe_dump({"per_file": all_results, "mean": mean}, f)

@@ -115,3 +115,37 @@ def test_dcn_backward_replay_determinism():
                                     1, 1, 1, 1, 1, 1, dg)[0]
     # fp32 atomics reorder rounding only: tight but not bitwise
     assert torch.allclose(g1, g2, atol=1e-5, rtol=1e-5)
+
+
+def test_graph_capture_preserves_weights():
+    """capture()'s warmup steps must not leave garbage-gradient Adam updates
+    in the model: params bit-identical before/after capture, optimizer state
+    fresh (advisor finding r1)."""
+    from esr_amd.engine.graph_runner import GraphedBPTTStep, flatten_grads
+    from esr_amd.models import build_model
+
+    device = torch.device("cuda:0")
+    torch.manual_seed(0)
+    model = build_model("ESRNet", inch=2, basech=8, num_frame=3,
+                        upsampler="pixelshuffle").to(device)
+    params = [p for p in model.parameters() if p.requires_grad]
+    before = [p.detach().clone() for p in params]
+    flat = flatten_grads(params, device)
+    opt = torch.optim.Adam(params, lr=1e-3, capturable=True)
+    runner = GraphedBPTTStep(
+        model, opt, flat, n_windows=2, inp_shape=(2, 4, 2, 32, 32),
+        gt_shape=(2, 2, 2, 32, 32), device=device,
+        amp_dtype=torch.bfloat16, sequence=True, seqn=3)
+    runner.capture()
+    for p, b in zip(params, before):
+        assert torch.equal(p.detach(), b), "capture() mutated model weights"
+    for state in opt.state.values():
+        for v in state.values():
+            if torch.is_tensor(v):
+                assert not v.any(), "capture() left non-fresh optimizer state"
+    # and the captured graph still trains: one replay changes the weights
+    x = torch.randn(2, 4, 2, 32, 32, device=device)
+    g = torch.randn(2, 2, 2, 32, 32, device=device)
+    runner.run([x], [g])
+    torch.cuda.synchronize()
+    assert any(not torch.equal(p.detach(), b) for p, b in zip(params, before))

@@ -164,3 +164,35 @@ def test_lpips_deterministic_and_discriminative():
     assert abs(p1(a, b).item() - p2(a, b).item()) < 1e-6
     assert p1(a, a).item() < 1e-6
     assert p1(a, b).item() > p1(a, a).item()
+
+
+def test_lpips_reference_heads_and_rng_isolation():
+    """The bundled v0.1 linear heads load by default and constructing the
+    metric neither reseeds nor advances the global torch RNG."""
+    from esr_amd.loss.lpips import LPIPS
+    torch.manual_seed(77)
+    before = torch.rand(4)
+    torch.manual_seed(77)
+    m = LPIPS("alex")
+    assert torch.equal(torch.rand(4), before)
+    assert m.heads_pretrained and not m.backbone_pretrained
+    # head channel widths match the alex slice dims (the reference v0.1 file)
+    assert [lin.weight.shape[1] for lin in m.lins] == [64, 192, 384, 256, 256]
+    assert all((lin.weight >= 0).all() for lin in m.lins)  # trained heads are nonneg
+
+
+def test_lpips_backbone_mapping_roundtrip():
+    from esr_amd.loss.lpips import LPIPS, _map_torchvision_backbone
+    g = torch.Generator().manual_seed(0)
+    shapes = [(64, 3, 11, 11), (192, 64, 5, 5), (384, 192, 3, 3),
+              (256, 384, 3, 3), (256, 256, 3, 3)]
+    sd = {}
+    for shp, idx in zip(shapes, [0, 3, 6, 8, 10]):  # torchvision alexnet layout
+        sd[f"features.{idx}.weight"] = torch.randn(*shp, generator=g)
+        sd[f"features.{idx}.bias"] = torch.randn(shp[0], generator=g)
+    m = LPIPS("alex")
+    m.features.load_state_dict(_map_torchvision_backbone("alex", sd))
+    assert torch.equal(m.features.slice1[0].weight, sd["features.0.weight"])
+    assert torch.equal(m.features.slice5[0].bias, sd["features.10.bias"])
+    m2 = LPIPS("alex")
+    assert not m2.backbone_pretrained
