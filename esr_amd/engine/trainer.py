@@ -102,6 +102,12 @@ class Trainer:
                            and self.device.type == "cuda")
         self._graph_step = None
 
+        # failure detection (the reference hangs forever on a
+        # dead rank); heartbeats in the training loop, monitor thread
+        # aborts so the launcher can relaunch from the last checkpoint
+        self._wd_cfg = tcfg.get("watchdog", {})
+        self.watchdog = None
+
         if resume is not None:
             self._resume_checkpoint(resume, reset)
 
@@ -229,10 +235,35 @@ class Trainer:
     # ---------------- loops ----------------
 
     def train(self):
-        if self.training_mode == "iteration_based_train":
-            self.iteration_based_training()
-        else:
-            self.epoch_based_training()
+        self._start_watchdog()
+        try:
+            if self.training_mode == "iteration_based_train":
+                self.iteration_based_training()
+            else:
+                self.epoch_based_training()
+        finally:
+            if self.watchdog is not None:
+                self.watchdog.stop()
+
+    def _start_watchdog(self):
+        import torch.distributed as dist
+        if not (dist.is_available() and dist.is_initialized()
+                and dist.get_world_size() > 1):
+            return
+        if not self._wd_cfg.get("enabled", True):
+            return
+        from ..parallel import Watchdog
+        self.watchdog = Watchdog(
+            timeout=float(self._wd_cfg.get("timeout", 120.0)),
+            interval=float(self._wd_cfg.get("interval", 5.0))).start()
+        if get_rank() == 0:
+            self.logger.info(
+                f"Watchdog armed: timeout {self.watchdog.timeout}s, "
+                f"interval {self.watchdog.interval}s")
+
+    def _beat(self):
+        if self.watchdog is not None:
+            self.watchdog.beat()
 
     def iteration_based_training(self):
         self.model.train()
@@ -248,6 +279,7 @@ class Trainer:
                     + self.start_iteration
                 best = False
                 loss, mse_loss, pred = self.bptt_step(inputs_seq, train=True)
+                self._beat()
 
                 reduced_mse = reduce_tensor(mse_loss)
                 reduced_loss = reduce_tensor(loss)
@@ -307,6 +339,7 @@ class Trainer:
             self.train_metrics.reset()
             for idx, inputs_seq in enumerate(self.train_dataloader):
                 loss, mse_loss, _ = self.bptt_step(inputs_seq, train=True)
+                self._beat()
                 reduced_mse = reduce_tensor(mse_loss)
                 reduced_loss = reduce_tensor(loss)
                 if get_rank() == 0:
