@@ -13,6 +13,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.conv import conv2d_act
 from ..ops.convgru import ConvGRUCell
 
 __all__ = [
@@ -53,9 +54,15 @@ class ConvLayer(nn.Module):
         self.conv2d = nn.Conv2d(in_channels, out_channels, kernel_size,
                                 stride, padding, bias=bias)
         self.activation = _ACTS[activation]
+        self._act_name = activation
         self.norm_layer = _norm_layer(norm, out_channels, bn_momentum)
 
     def forward(self, x):
+        if self.norm_layer is None:
+            # native gfx950 conv with bias+act fused into the epilogue
+            out = conv2d_act(x, self.conv2d, self._act_name)
+            if out is not None:
+                return out
         out = self.conv2d(x)
         if self.norm_layer is not None:
             out = self.norm_layer(out)
@@ -80,14 +87,19 @@ class ResidualBlock(nn.Module):
 
     def forward(self, x):
         residual = x if self.downsample is None else self.downsample(x)
-        out = self.conv1(x)
-        if self.n1 is not None:
-            out = self.n1(out)
-        out = F.relu(out)
-        out = self.conv2(out)
-        if self.n2 is not None:
-            out = self.n2(out)
-        out = out + residual
+        if self.n1 is None:
+            out = conv2d_act(x, self.conv1, "relu")
+            if out is None:
+                out = F.relu(self.conv1(x))
+        else:
+            out = F.relu(self.n1(self.conv1(x)))
+        if self.n2 is None:
+            out2 = conv2d_act(out, self.conv2, None)
+            if out2 is None:
+                out2 = self.conv2(out)
+        else:
+            out2 = self.n2(self.conv2(out))
+        out = out2 + residual
         if self.final_activation:
             out = F.relu(out)
         return out
@@ -103,12 +115,17 @@ class UpsampleConvLayer(nn.Module):
         self.conv2d = nn.Conv2d(in_channels, out_channels, kernel_size,
                                 stride, padding, bias=bias)
         self.activation = _ACTS[activation]
+        self._act_name = activation
         self.norm_layer = _norm_layer(norm, out_channels)
         self.scale = scale
 
     def forward(self, x):
         x = F.interpolate(x, scale_factor=self.scale, mode="bilinear",
                           align_corners=False)
+        if self.norm_layer is None:
+            out = conv2d_act(x, self.conv2d, self._act_name)
+            if out is not None:
+                return out
         out = self.conv2d(x)
         if self.norm_layer is not None:
             out = self.norm_layer(out)
@@ -149,7 +166,10 @@ class PixelShuffleUpsample(nn.Module):
             sub.repeat_interleave(r2, dim=0).reshape_as(w))
 
     def forward(self, x):
-        out = F.pixel_shuffle(self.conv2d(x), self.scale)
+        pre = conv2d_act(x, self.conv2d, None)
+        if pre is None:
+            pre = self.conv2d(x)
+        out = F.pixel_shuffle(pre, self.scale)
         if self.norm_layer is not None:
             out = self.norm_layer(out)
         if self.activation is not None:
@@ -196,7 +216,10 @@ class ConvLSTMCell(nn.Module):
                             dtype=x.dtype, device=x.device)
             state = (z, z)
         h_prev, c_prev = state
-        g = self.gates(torch.cat([x, h_prev], dim=1))
+        xh = torch.cat([x, h_prev], dim=1)
+        g = conv2d_act(xh, self.gates, None)
+        if g is None:
+            g = self.gates(xh)
         i, f, o, c_hat = torch.chunk(g, 4, dim=1)
         i, f, o = torch.sigmoid(i), torch.sigmoid(f), torch.sigmoid(o)
         c = f * c_prev + i * torch.tanh(c_hat)

@@ -1,0 +1,143 @@
+"""Native bf16 NCHW convolution dispatch (hand-written gfx950 kernels).
+
+Replaces MIOpen + NCHW<->NHWC transposes + autocast casts on the conv
+shapes ESRNet runs (reference ConvLayer ESR:models/submodules.py:159-200,
+ConvGRU convs :474-514): 3x3 stride-1/2 pad-1 and 1x1 convs, bias and
+ReLU/sigmoid/tanh fused into the conv epilogue.
+
+Dispatch (see conv2d.hip):
+  * MFMA implicit-GEMM kernel for Cout >= 32 (the compute-heavy deep
+    convs at 32x32/64x64);
+  * direct VALU kernel for the bandwidth-bound small-channel shapes
+    (head/tail/attention convs).
+Backward: stride-1 input-grad reuses the forward kernels with
+flipped/transposed packed weights; stride-2 input-grad and the MFMA
+split-K weight-grad are dedicated kernels.  fp32 falls back to torch
+(the CPU oracle used by the parity tests).
+
+Env: ESR_NATIVE_CONV=0 disables the native path (A/B benching).
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+import torch.nn.functional as F
+
+from .native import get_ext
+
+__all__ = ["conv2d_act", "native_conv_supported", "ACT_IDS"]
+
+ACT_IDS = {None: 0, "none": 0, "relu": 1, "sigmoid": 2, "tanh": 3}
+
+_MFMA_MIN_COUT = 32   # below this the MFMA M-tile would idle; use VALU
+_CIK = 32             # K-chunk of the MFMA kernel (pad Cin up to this)
+
+
+def _enabled() -> bool:
+    return os.environ.get("ESR_NATIVE_CONV", "1") != "0"
+
+
+def _ceil(v: int, m: int) -> int:
+    return (v + m - 1) // m * m
+
+
+def _pack(w: torch.Tensor) -> torch.Tensor:
+    """[O, I, k, k] -> [k*k, O_p, I_p] bf16, O_p = ceil16(O), I_p = ceil32(I).
+
+    Matches the A-fragment addressing of conv2d_fwd_mfma_kernel: row = cout,
+    contiguous 8-channel groups along ci.
+    """
+    O, I, k, _ = w.shape
+    t = w.permute(2, 3, 0, 1).reshape(k * k, O, I)
+    op, ip = _ceil(O, 16), _ceil(I, _CIK)
+    if op != O or ip != I:
+        t = F.pad(t, (0, ip - I, 0, op - O))
+    return t.contiguous()
+
+
+class _NativeConv2dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, act_id):
+        ext = get_ext()
+        ks = w.shape[2]
+        cout = w.shape[0]
+        bias_f = bias.float() if bias is not None else None
+        if cout >= _MFMA_MIN_COUT:
+            y = ext.conv2d_fwd_mfma(x, _pack(w), bias_f, cout, ks, stride,
+                                    act_id)
+        else:
+            y = ext.conv2d_fwd_valu(x, w.contiguous(), bias_f, stride, act_id)
+        ctx.save_for_backward(x, w, y)
+        ctx.stride_ = stride
+        ctx.act_id = act_id
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext()
+        x, w, y = ctx.saved_tensors
+        stride, act_id = ctx.stride_, ctx.act_id
+        ks = w.shape[2]
+        cout, cin = w.shape[0], w.shape[1]
+        dy = dy.contiguous()
+        dpre = ext.act_grad(dy, y, act_id) if act_id else dy
+
+        # input grad: stride-1 is this conv with flipped/transposed weights
+        if stride == 1:
+            wt = w.transpose(0, 1)
+            if ks == 3:
+                wt = wt.flip((2, 3))
+            if cin >= _MFMA_MIN_COUT:
+                dx = ext.conv2d_fwd_mfma(dpre, _pack(wt), None, cin, ks, 1, 0)
+            else:
+                dx = ext.conv2d_fwd_valu(dpre, wt.contiguous(), None, 1, 0)
+        else:
+            dx = ext.conv2d_dgrad_s2(dpre, w.contiguous(),
+                                     x.shape[2], x.shape[3])
+
+        # weight grad: MFMA split-K into padded fp32, then slice
+        dwp = ext.conv2d_wgrad_mfma(x, dpre, ks, stride,
+                                    _ceil(cin, 16), _ceil(cout, 16))
+        dw = dwp[:cout, :cin].to(w.dtype)
+        db = dpre.sum(dim=(0, 2, 3), dtype=torch.float32).to(w.dtype) \
+            if ctx.has_bias else None
+        return dx, dw, db, None, None
+
+
+def native_conv_supported(x: torch.Tensor, conv: torch.nn.Conv2d) -> bool:
+    if not (_enabled() and x.is_cuda and get_ext() is not None):
+        return False
+    if conv.groups != 1 or conv.dilation != (1, 1):
+        return False
+    kh, kw = conv.kernel_size
+    sh, sw = conv.stride
+    ph, pw = conv.padding if isinstance(conv.padding, tuple) else \
+        (conv.padding, conv.padding)
+    if kh != kw or sh != sw or ph != pw:
+        return False
+    if (kh, sh) not in ((3, 1), (3, 2), (1, 1)):
+        return False
+    if ph != kh // 2:
+        return False
+    return True
+
+
+def conv2d_act(x: torch.Tensor, conv: torch.nn.Conv2d, act: str | None):
+    """Fused conv+bias+act.  Returns the native-kernel result when the shape
+    and dtype qualify, else None (caller falls back to torch)."""
+    if act not in ACT_IDS or not native_conv_supported(x, conv):
+        return None
+    if torch.is_autocast_enabled():
+        x = x.to(torch.bfloat16)
+    if x.dtype != torch.bfloat16:
+        return None
+    w, b = conv.weight, conv.bias
+    if w.dtype != torch.bfloat16:
+        w = w.to(torch.bfloat16)          # autograd records the cast:
+        if b is not None:                  # grads flow back to fp32 masters
+            b = b.to(torch.bfloat16)
+    return _NativeConv2dFn.apply(x.contiguous(), w, b, conv.stride[0],
+                                 ACT_IDS[act])

@@ -20,6 +20,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
+from .conv import conv2d_act
 from .native import get_ext, require_ext
 
 __all__ = ["ConvGRUCell", "gru_gates_ur", "gru_gates_out"]
@@ -115,7 +116,12 @@ class ConvGRUCell(nn.Module):
             h = torch.zeros(x.size(0), self.hidden_size, x.size(2), x.size(3),
                             dtype=x.dtype, device=x.device)
         xh = torch.cat([x, h], dim=1)
-        ur_pre = self.ur_gate(xh)
+        ur_pre = conv2d_act(xh, self.ur_gate, None)
+        if ur_pre is None:
+            ur_pre = self.ur_gate(xh)
         u, r, hr = gru_gates_ur(ur_pre, h)
-        o_pre = self.out_gate(torch.cat([x, hr], dim=1))
+        xhr = torch.cat([x, hr], dim=1)
+        o_pre = conv2d_act(xhr, self.out_gate, None)
+        if o_pre is None:
+            o_pre = self.out_gate(xhr)
         return gru_gates_out(o_pre, u, h)

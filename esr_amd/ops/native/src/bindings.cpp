@@ -36,6 +36,19 @@ at::Tensor splat_count(const at::Tensor&, int64_t, int64_t);
 at::Tensor splat_stack(const at::Tensor&, int64_t, int64_t, int64_t, double,
                        double);
 
+// conv2d.hip
+at::Tensor conv2d_fwd_mfma(const at::Tensor&, const at::Tensor&,
+                           const c10::optional<at::Tensor>&,
+                           int64_t, int64_t, int64_t, int64_t);
+at::Tensor conv2d_fwd_valu(const at::Tensor&, const at::Tensor&,
+                           const c10::optional<at::Tensor>&, int64_t, int64_t);
+at::Tensor conv2d_dgrad_s2(const at::Tensor&, const at::Tensor&,
+                           int64_t, int64_t);
+at::Tensor conv2d_wgrad_mfma(const at::Tensor&, const at::Tensor&,
+                             int64_t, int64_t, int64_t, int64_t);
+at::Tensor act_grad(const at::Tensor&, const at::Tensor&, int64_t);
+at::Tensor gemm16_probe(const at::Tensor&, const at::Tensor&);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("deform_conv2d_forward", &deform_conv2d_forward,
         "modulated deformable conv forward (gfx950)");
@@ -51,5 +64,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gru_gates_out_backward", &gru_gates_out_backward);
   m.def("splat_count", &splat_count);
   m.def("splat_stack", &splat_stack);
+  m.def("conv2d_fwd_mfma", &conv2d_fwd_mfma,
+        "bf16 NCHW conv fwd, MFMA implicit GEMM, fused bias+act (gfx950)");
+  m.def("conv2d_fwd_valu", &conv2d_fwd_valu,
+        "bf16 NCHW conv fwd, direct VALU, fused bias+act (gfx950)");
+  m.def("conv2d_dgrad_s2", &conv2d_dgrad_s2,
+        "bf16 stride-2 conv input-grad (gfx950)");
+  m.def("conv2d_wgrad_mfma", &conv2d_wgrad_mfma,
+        "bf16 conv weight-grad, MFMA split-K + fp32 atomics (gfx950)");
+  m.def("act_grad", &act_grad, "fused activation backward (bf16)");
+  m.def("gemm16_probe", &gemm16_probe, "16x16x32 bf16 MFMA fragment probe");
   m.attr("gfx_arch") = "gfx950";
 }
