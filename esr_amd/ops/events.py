@@ -210,7 +210,8 @@ def _expand_linspace(counts: torch.Tensor, t0: torch.Tensor, t1: torch.Tensor,
 
 
 def redistribute_stack(stack: torch.Tensor, mode: str = "linear",
-                       generator=None) -> torch.Tensor:
+                       generator=None, capacity: int | None = None
+                       ) -> torch.Tensor:
     """Count stack -> event cloud, the inverse of the splatting ops.
 
     Input: [B, C, Y, X] (no-polarity) or [B, P, C, Y, X] (polarity) count
@@ -221,9 +222,16 @@ def redistribute_stack(stack: torch.Tensor, mode: str = "linear",
     and ESR:dataloader/encodings.py:366-463: a cell with value v at bin c
     emits |v| events at that pixel with timestamps in
     (c/C + 1/(100C), (c+1)/C], polarity sign(v); per-item global sort by t.
-    This implementation is fully vectorized (repeat_interleave + stable sort)
-    and therefore also runs on GPU tensors.
+
+    On GPU with the native extension this runs as one device pipeline
+    (prefix scan + scatter + segmented radix sort, redistribute.hip) with
+    no host syncs when ``capacity`` (max events per item) is given —
+    graph-capturable; without ``capacity`` one sync computes the true max
+    length.  The vectorized torch path below is the CPU oracle.
     """
+    native = _native_redistribute(stack, mode, generator, capacity)
+    if native is not None:
+        return native
     if stack.dim() == 5:
         Bb, P, C, Y, X = stack.shape
         s = stack.round().reshape(Bb, P * C, Y, X)
@@ -267,8 +275,38 @@ def redistribute_stack(stack: torch.Tensor, mode: str = "linear",
     return out
 
 
+def _native_redistribute(stack, mode, generator, capacity):
+    """Device pipeline dispatch; returns None when the torch path applies."""
+    if not stack.is_cuda or mode not in ("linear", "random"):
+        return None
+    from .native import get_ext
+    ext = get_ext()
+    if ext is None or not hasattr(ext, "redistribute_stack_hip"):
+        return None
+    if stack.dim() == 5:
+        Bb, P, C, Y, X = stack.shape
+        flat = stack.reshape(Bb, P * C, Y, X)
+    elif stack.dim() == 4:
+        C = stack.size(1)
+        flat = stack
+    else:
+        raise ValueError("stack must be 4D [B,C,Y,X] or 5D [B,P,C,Y,X]")
+    flat = flat.float().contiguous()
+    sync_trim = capacity is None
+    if sync_trim:
+        # one sync for the true max length (the Cython parity shape);
+        # pass capacity explicitly for the graph-capturable path
+        capacity = max(1, int(flat.round().abs().sum(dim=(1, 2, 3)).max()))
+    seed = 123 if generator is None else \
+        int(generator.initial_seed()) & 0xFFFFFFFF
+    events, lengths = ext.redistribute_stack_hip(
+        flat, C, int(capacity), 0 if mode == "linear" else 1, seed)
+    return events
+
+
 def redistribute_count(cnt: torch.Tensor, mode: str = "linear",
-                       generator=None) -> torch.Tensor:
+                       generator=None, capacity: int | None = None
+                       ) -> torch.Tensor:
     """2-channel count map [B,2,H,W] -> event cloud [B,N,4].
 
     Parity: ESR:dataloader/cython_cnt2event/cnt2event.pyx:18-116.  A count map
@@ -277,7 +315,8 @@ def redistribute_count(cnt: torch.Tensor, mode: str = "linear",
     if cnt.dim() != 4 or cnt.size(1) != 2:
         raise ValueError("cnt must be [B,2,H,W]")
     signed = torch.stack([cnt[:, 0], -cnt[:, 1]], dim=1)   # [B,2,H,W]
-    return redistribute_stack(signed.unsqueeze(2), mode=mode, generator=generator)
+    return redistribute_stack(signed.unsqueeze(2), mode=mode,
+                              generator=generator, capacity=capacity)
 
 
 def event_formatting(events) -> torch.Tensor:

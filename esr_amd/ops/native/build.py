@@ -19,7 +19,7 @@ HERE = Path(__file__).resolve().parent
 SRC = HERE / "src"
 SOURCES = [SRC / "bindings.cpp", SRC / "deform_conv.hip",
            SRC / "deform_conv_fused.hip", SRC / "gru_gates.hip",
-           SRC / "event_ops.hip", SRC / "conv2d.hip"]
+           SRC / "event_ops.hip", SRC / "conv2d.hip", SRC / "redistribute.hip"]
 
 
 def build(verbose: bool = False) -> str:

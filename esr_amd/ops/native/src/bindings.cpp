@@ -49,6 +49,10 @@ at::Tensor conv2d_wgrad_mfma(const at::Tensor&, const at::Tensor&,
 at::Tensor act_grad(const at::Tensor&, const at::Tensor&, int64_t);
 at::Tensor gemm16_probe(const at::Tensor&, const at::Tensor&);
 
+// redistribute.hip
+std::vector<at::Tensor> redistribute_stack_hip(const at::Tensor&, int64_t,
+                                               int64_t, int64_t, int64_t);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("deform_conv2d_forward", &deform_conv2d_forward,
         "modulated deformable conv forward (gfx950)");
@@ -74,5 +78,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "bf16 conv weight-grad, MFMA split-K + fp32 atomics (gfx950)");
   m.def("act_grad", &act_grad, "fused activation backward (bf16)");
   m.def("gemm16_probe", &gemm16_probe, "16x16x32 bf16 MFMA fragment probe");
+  m.def("redistribute_stack_hip", &redistribute_stack_hip,
+        "count stack -> sorted event cloud, device-only "
+        "(scan + scatter + segmented radix sort)");
   m.attr("gfx_arch") = "gfx950";
 }

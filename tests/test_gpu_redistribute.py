@@ -1,0 +1,119 @@
+"""GPU tests for the device-only count->event redistribution pipeline
+(redistribute.hip: scan + scatter + segmented radix sort)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def _rand_stack(shape, seed=0, lo=-3, hi=4):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randint(lo, hi, shape, generator=g).float()
+
+
+def test_redistribute_gpu_matches_cpu_oracle():
+    """Same stack through the HIP pipeline and the vectorized torch path
+    (the CPU oracle): identical events per item (linear mode)."""
+    from esr_amd.ops.events import redistribute_stack
+    stack = _rand_stack((3, 2, 4, 16, 16), seed=7)
+    cpu = redistribute_stack(stack, mode="linear")
+    gpu = redistribute_stack(stack.to(DEV), mode="linear").cpu()
+    assert gpu.shape == cpu.shape, (gpu.shape, cpu.shape)
+    # events with distinct t must match exactly; equal-t runs may permute
+    # (stable orders agree here by construction, so compare directly with a
+    # small t tolerance)
+    assert torch.allclose(gpu[..., 2], cpu[..., 2], atol=1e-5)
+    # per-pixel/per-polarity histograms are exactly equal
+    for b in range(stack.size(0)):
+        for pol in (1.0, -1.0):
+            mc = cpu[b][(cpu[b, :, 3] == pol)]
+            mg = gpu[b][(gpu[b, :, 3] == pol)]
+            hc = torch.zeros(16, 16)
+            hg = torch.zeros(16, 16)
+            hc.index_put_((mc[:, 1].long(), mc[:, 0].long()),
+                          torch.ones(mc.size(0)), accumulate=True)
+            hg.index_put_((mg[:, 1].long(), mg[:, 0].long()),
+                          torch.ones(mg.size(0)), accumulate=True)
+            assert torch.equal(hc, hg)
+
+
+def test_redistribute_gpu_sorted_and_padded():
+    from esr_amd.ops.events import redistribute_stack
+    stack = _rand_stack((2, 3, 8, 8), seed=3).to(DEV)
+    ev = redistribute_stack(stack, mode="linear")
+    n_true = [int(stack[b].round().abs().sum()) for b in range(2)]
+    assert ev.size(1) == max(n_true)
+    for b in range(2):
+        t = ev[b, : n_true[b], 2]
+        assert (t[1:] >= t[:-1]).all(), "not time-sorted"
+        assert (ev[b, n_true[b]:] == 0).all(), "padding not zero"
+        assert ((ev[b, : n_true[b], 3].abs() == 1).all())
+
+
+def test_redistribute_gpu_capacity_and_overflow():
+    from esr_amd.ops.events import redistribute_stack
+    stack = _rand_stack((2, 1, 8, 8), seed=9, lo=0, hi=5).to(DEV)
+    full = redistribute_stack(stack, mode="linear")
+    n0 = int(stack[0].round().abs().sum())
+    cap = max(4, n0 // 2)
+    ev = redistribute_stack(stack, mode="linear", capacity=cap)
+    assert ev.shape[1] == cap
+    # truncation keeps a sorted prefix of the full stream
+    kept = ev[0, :, 2]
+    kept = kept[kept > 0]
+    ref = full[0, : n0, 2]
+    assert torch.allclose(kept, ref[: kept.numel()], atol=1e-6)
+
+
+def test_redistribute_count_roundtrip_gpu():
+    """count map -> events -> splat back == original (the reference's own
+    round-trip invariant, ESR:dataloader/encodings.py:673-696)."""
+    from esr_amd.ops.events import redistribute_count
+    from esr_amd.ops.native import require_ext
+    ext = require_ext()
+    cnt = _rand_stack((4, 2, 32, 32), seed=11, lo=0, hi=4).to(DEV)
+    ev = redistribute_count(cnt, mode="linear")
+    back = ext.splat_count(ev.contiguous(), 32, 32)
+    assert torch.equal(back, cnt), "round-trip changed the count map"
+
+
+def test_redistribute_gpu_graph_capturable():
+    """With explicit capacity the pipeline records into a hipGraph."""
+    from esr_amd.ops.events import redistribute_stack
+    stack = _rand_stack((2, 2, 16, 16), seed=5, lo=0, hi=3).to(DEV)
+    cap = 1024
+    # warmup on side stream
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(2):
+            redistribute_stack(stack, mode="linear", capacity=cap)
+    torch.cuda.current_stream().wait_stream(s)
+    torch.cuda.synchronize()
+    out = [None]
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        out[0] = redistribute_stack(stack, mode="linear", capacity=cap)
+    torch.cuda.synchronize()
+    ref = redistribute_stack(stack, mode="linear", capacity=cap).clone()
+    stackv = stack  # static input; replay recomputes into out[0]
+    out[0].zero_()
+    g.replay()
+    torch.cuda.synchronize()
+    assert torch.allclose(out[0], ref)
+
+
+def test_redistribute_gpu_random_mode_bounds():
+    from esr_amd.ops.events import redistribute_stack
+    stack = _rand_stack((2, 4, 8, 8), seed=13).to(DEV)
+    ev = redistribute_stack(stack, mode="random")
+    n0 = int(stack[0].round().abs().sum())
+    t = ev[0, :n0, 2]
+    assert (t > 0).all() and (t <= 1.0).all()
+    assert (t[1:] >= t[:-1]).all()
+    # deterministic for the same (default) seed
+    ev2 = redistribute_stack(stack, mode="random")
+    assert torch.equal(ev, ev2)

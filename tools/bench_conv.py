@@ -9,10 +9,14 @@ Usage (GPU box): python tools/bench_conv.py [--iters 50]
 """
 
 import argparse
+import sys
 import time
+from pathlib import Path
 
 import torch
 import torch.nn.functional as F
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 # (label, frames, Cin, Cout, H, W, ks, stride, act) — flagship step shapes
 SHAPES = [
