@@ -5,17 +5,25 @@ shapes ESRNet runs (reference ConvLayer ESR:models/submodules.py:159-200,
 ConvGRU convs :474-514): 3x3 stride-1/2 pad-1 and 1x1 convs, bias and
 ReLU/sigmoid/tanh fused into the conv epilogue.
 
-Dispatch (see conv2d.hip):
-  * MFMA implicit-GEMM kernel for Cout >= 32 (the compute-heavy deep
-    convs at 32x32/64x64);
-  * direct VALU kernel for the bandwidth-bound small-channel shapes
-    (head/tail/attention convs).
-Backward: stride-1 input-grad reuses the forward kernels with
-flipped/transposed packed weights; stride-2 input-grad and the MFMA
-split-K weight-grad are dedicated kernels.  fp32 falls back to torch
-(the CPU oracle used by the parity tests).
+Dispatch (see conv2d.hip; thresholds from tools/bench_conv.py measured on
+MI355X, profiles/README.md):
+  * MFMA implicit-GEMM kernel for Cout >= 32 — measured 1.0-2.0x MIOpen
+    bf16 on the deep/mid shapes;
+  * direct VALU kernel for Cin >= 32 with tiny Cout (attention/kernel
+    convs, 1.2-1.3x);
+  * shapes with BOTH Cin < 32 and Cout < 32 (head/enc1/tail) fall back to
+    torch/MIOpen — the v1 VALU kernel measured 0.2-0.7x there (scalar
+    loads + per-tap branches; an instruction-count-optimized v2 is the
+    open item).
+Backward defaults to torch's aten.convolution_backward (MIOpen bf16):
+the native stride-1 dgrad (= this forward with flipped/transposed packed
+weights), stride-2 dgrad and MFMA split-K wgrad kernels are implemented
+and oracle-tested but measured slower than MIOpen at v1
+(ESR_CONV_BWD=native selects them; bench_conv.py --bwd has numbers).
+fp32 falls back to torch everywhere (the CPU oracle of the parity tests).
 
-Env: ESR_NATIVE_CONV=0 disables the native path (A/B benching).
+Env: ESR_NATIVE_CONV=0 disables the native path (A/B benching);
+     ESR_CONV_BWD=native routes backward to the native kernels.
 """
 
 from __future__ import annotations
@@ -62,7 +70,7 @@ class _NativeConv2dFn(torch.autograd.Function):
     def forward(ctx, x, w, bias, stride, act_id):
         ext = get_ext()
         ks = w.shape[2]
-        cout = w.shape[0]
+        cout, cin = w.shape[0], w.shape[1]
         bias_f = bias.float() if bias is not None else None
         if cout >= _MFMA_MIN_COUT:
             y = ext.conv2d_fwd_mfma(x, _pack(w), bias_f, cout, ks, stride,
@@ -85,6 +93,14 @@ class _NativeConv2dFn(torch.autograd.Function):
         dy = dy.contiguous()
         dpre = ext.act_grad(dy, y, act_id) if act_id else dy
 
+        if os.environ.get("ESR_CONV_BWD", "aten") != "native":
+            dx, dw, db = torch.ops.aten.convolution_backward(
+                dpre, x, w, [cout] if ctx.has_bias else None,
+                [stride, stride], [ks // 2, ks // 2], [1, 1], False, [0, 0],
+                1, [True, True, ctx.has_bias])
+            return dx, dw, db if ctx.has_bias else None, None, None
+
+        # --- native backward path (ESR_CONV_BWD=native) ---
         # input grad: stride-1 is this conv with flipped/transposed weights
         if stride == 1:
             wt = w.transpose(0, 1)
@@ -109,6 +125,10 @@ class _NativeConv2dFn(torch.autograd.Function):
 
 def native_conv_supported(x: torch.Tensor, conv: torch.nn.Conv2d) -> bool:
     if not (_enabled() and x.is_cuda and get_ext() is not None):
+        return False
+    if conv.out_channels < _MFMA_MIN_COUT and conv.in_channels < _MFMA_MIN_COUT:
+        # head/enc1/tail class: MIOpen measured faster than the v1 VALU
+        # kernel on these bandwidth-bound shapes (tools/bench_conv.py)
         return False
     if conv.groups != 1 or conv.dilation != (1, 1):
         return False

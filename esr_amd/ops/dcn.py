@@ -199,7 +199,10 @@ class DeformAlign2d(nn.Module):
         # (ESR:models/DCNv2/dcn_v2.py:218-219) only permutes learned
         # channels, so skipping it is math-equivalent under training and
         # saves a concat launch per call.
-        om = self.conv_offset_mask(feat)
+        from .conv import conv2d_act
+        om = conv2d_act(feat, self.conv_offset_mask, None)
+        if om is None:
+            om = self.conv_offset_mask(feat)
         kh, kw = self.kernel_size
         n_off = self.deformable_groups * 2 * kh * kw
         offset = om[:, :n_off]

@@ -99,6 +99,7 @@ def test_conv_forward_matches_oracle(cin, cout, h, w, ks, stride, act):
     _assert_close(y, ref, what=f"fwd {cin}->{cout} k{ks}s{stride} {act}")
 
 
+@pytest.mark.parametrize("bwd", ["aten", "native"])
 @pytest.mark.parametrize("cin,cout,h,w,ks,stride,act", [
     (2, 8, 64, 64, 3, 1, "relu"),
     (32, 64, 32, 32, 3, 2, "relu"),
@@ -109,7 +110,9 @@ def test_conv_forward_matches_oracle(cin, cout, h, w, ks, stride, act):
     (40, 48, 30, 30, 3, 1, "tanh"),
     (8, 16, 64, 64, 3, 2, "relu"),
 ])
-def test_conv_backward_matches_oracle(cin, cout, h, w, ks, stride, act):
+def test_conv_backward_matches_oracle(cin, cout, h, w, ks, stride, act, bwd,
+                                      monkeypatch):
+    monkeypatch.setenv("ESR_CONV_BWD", bwd)
     from esr_amd.ops.conv import ACT_IDS, _NativeConv2dFn
     x = _rand_bf16(3, cin, h, w, seed=cin * 3 + cout)
     wt = _rand_bf16(cout, cin, ks, ks, seed=cin - cout, scale=0.3)
