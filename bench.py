@@ -78,10 +78,17 @@ def main():
     p.add_argument("--seqn", type=int, default=3)
     p.add_argument("--window", type=int, default=2048)
     p.add_argument("--lr-size", type=int, default=128)
+    p.add_argument("--lr-h", type=int, default=None,
+                   help="LR height (defaults to --lr-size; for DVS-native "
+                        "resolutions like 180x240)")
+    p.add_argument("--lr-w", type=int, default=None)
     p.add_argument("--scale", type=int, default=2)
     p.add_argument("--basech", type=int, default=8)
     p.add_argument("--dtype", type=str, default="bf16",
-                   choices=["bf16", "fp32"])
+                   choices=["bf16", "fp16", "fp32"])
+    p.add_argument("--metric-suffix", type=str, default=None,
+                   help="override the config label in the JSON line "
+                        "(BASELINE config arms)")
     p.add_argument("--upsampler", type=str, default="pixelshuffle")
     p.add_argument("--no-graphs", action="store_true",
                    help="disable hipGraph capture (eager mode)")
@@ -153,13 +160,16 @@ def main():
                                      amsgrad=True, foreach=True,
                                      capturable=not args.no_graphs)
 
-    lr_res = (args.lr_size, args.lr_size)
-    hr_res = (args.lr_size * args.scale, args.lr_size * args.scale)
+    lr_h = args.lr_h or args.lr_size
+    lr_w = args.lr_w or args.lr_size
+    lr_res = (lr_h, lr_w)
+    hr_res = (lr_h * args.scale, lr_w * args.scale)
     n_windows = args.seql - args.seqn + 1
     pools = make_sequence_pool(2, args.seql, n_windows, args.batch,
                                args.seqn, args.window, lr_res, hr_res,
                                device, ext, seed=100 + rank)
-    amp_dtype = torch.bfloat16 if args.dtype == "bf16" else None
+    amp_dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
+                 "fp32": None}[args.dtype]
 
     # graph replay reads from fixed addresses: one graph per pre-generated
     # data set (no per-step copies), all sharing one memory pool
@@ -346,9 +356,11 @@ def main():
     value = total_frames / elapsed
     ms_per_step = elapsed / args.steps * 1e3
 
+    metric_name = args.metric_suffix or \
+        f"SR event-frames/sec (whole node), {args.scale}x NFS-syn"
     if rank == 0:
         print(json.dumps({
-            "metric": "SR event-frames/sec (whole node), 2x NFS-syn",
+            "metric": metric_name,
             "value": round(value, 2),
             "unit": "frames/s",
             "n_gpus": world,
@@ -356,6 +368,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(ms_per_step, 2),
             "higher_is_better": True,
+            "peak_mem_gb": round(torch.cuda.max_memory_allocated() / 2**30, 2),
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": args.dtype,
@@ -366,7 +379,7 @@ def main():
                 "seq_len": args.seql,
                 "seqn": args.seqn,
                 "window_events": args.window,
-                "input": f"{args.lr_size}->{args.lr_size * args.scale}",
+                "input": f"{lr_h}x{lr_w}->{lr_h * args.scale}x{lr_w * args.scale}",
                 "scale": args.scale,
                 "upsampler": args.upsampler,
                 "hip_graphs": graph_mode,
