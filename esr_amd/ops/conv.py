@@ -39,6 +39,9 @@ __all__ = ["conv2d_act", "native_conv_supported", "ACT_IDS"]
 
 ACT_IDS = {None: 0, "none": 0, "relu": 1, "sigmoid": 2, "tanh": 3}
 
+# dispatch telemetry: tests / profiling assert the native path really runs
+stats = {"native_calls": 0}
+
 _MFMA_MIN_COUT = 32   # below this the MFMA M-tile would idle; use VALU
 _CIK = 32             # K-chunk of the MFMA kernel (pad Cin up to this)
 
@@ -159,5 +162,6 @@ def conv2d_act(x: torch.Tensor, conv: torch.nn.Conv2d, act: str | None):
         w = w.to(torch.bfloat16)          # autograd records the cast:
         if b is not None:                  # grads flow back to fp32 masters
             b = b.to(torch.bfloat16)
+    stats["native_calls"] += 1
     return _NativeConv2dFn.apply(x.contiguous(), w, b, conv.stride[0],
                                  ACT_IDS[act])

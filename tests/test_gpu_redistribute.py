@@ -61,11 +61,15 @@ def test_redistribute_gpu_capacity_and_overflow():
     cap = max(4, n0 // 2)
     ev = redistribute_stack(stack, mode="linear", capacity=cap)
     assert ev.shape[1] == cap
-    # truncation keeps a sorted prefix of the full stream
-    kept = ev[0, :, 2]
-    kept = kept[kept > 0]
-    ref = full[0, : n0, 2]
-    assert torch.allclose(kept, ref[: kept.numel()], atol=1e-6)
+    # truncation drops whole-cell tails in cell-scan order (not time
+    # order); what must hold: exactly `cap` kept events, time-sorted, and
+    # every kept event present in the full stream
+    kept = ev[0][ev[0, :, 2] > 0]
+    assert kept.size(0) == cap
+    t = kept[:, 2]
+    assert (t[1:] >= t[:-1]).all()
+    full_set = {tuple(r) for r in full[0, :n0].tolist()}
+    assert all(tuple(r) in full_set for r in kept.tolist())
 
 
 def test_redistribute_count_roundtrip_gpu():
