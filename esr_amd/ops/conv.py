@@ -96,7 +96,10 @@ class _NativeConv2dFn(torch.autograd.Function):
         dy = dy.contiguous()
         dpre = ext.act_grad(dy, y, act_id) if act_id else dy
 
-        if os.environ.get("ESR_CONV_BWD", "aten") != "native":
+        mode = os.environ.get("ESR_CONV_BWD", "auto")
+        native_bwd = mode == "native" or (
+            mode == "auto" and stride == 1 and cin >= _MFMA_MIN_COUT)
+        if not native_bwd:
             dx, dw, db = torch.ops.aten.convolution_backward(
                 dpre, x, w, [cout] if ctx.has_bias else None,
                 [stride, stride], [ks // 2, ks // 2], [1, 1], False, [0, 0],

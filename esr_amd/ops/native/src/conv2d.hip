@@ -387,6 +387,144 @@ void conv2d_wgrad_mfma_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// MFMA weight-grad v2, stride-1 (3x3 p1 and 1x1): block-wide staged tiles,
+// all staging as aligned b128 vectors, kx-shifts resolved by REGISTER
+// shuffles of adjacent aligned fragments (no shifted LDS copies, no
+// unaligned ds_read — guide G17).
+//
+//   block tile: [64 co] x [32 ci] x taps;  wave w owns co slice w*16.
+//   A = dpre rows (halo'd u-domain), B = X rows; D[row=co][col=ci].
+//   K = output pixels, chunked 32 per output row; fp32 atomics out.
+// ---------------------------------------------------------------------------
+
+constexpr int WGV2_DPW = 56;  // dp row stride in shorts (112 B: 28-dword
+                              // stride, gcd(28,64)=4 -> conflict-free b128)
+constexpr int WGV2_XW = 40;   // x row stride (same bank math as SLOT)
+
+template <int KS>
+__global__ __launch_bounds__(256)
+void conv2d_wgrad_s1_kernel(
+    const ushort* __restrict__ x,     // [B, Cin, H, W]
+    const ushort* __restrict__ dpre,  // [B, Cout, H, W] (stride 1: Ho=H)
+    float* __restrict__ dwp,          // [Cout_p, Cin_p, KS, KS] fp32
+    int Cin, int H, int W, int Cout,
+    int Cin_p, int Cout_p, int uw) {
+  constexpr int PAD = KS / 2;
+  constexpr int NTAP = KS * KS;
+  __shared__ ushort dp[64 * WGV2_DPW];        // px domain [-8, 48)
+  __shared__ ushort xs[32 * KS * WGV2_XW];    // px domain [0, 32)
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  const int b = blockIdx.x / uw;
+  const int ux0 = (blockIdx.x % uw) * TW;
+  const int ci0 = blockIdx.y * 32;
+  const int co0 = blockIdx.z * 64;
+
+  const long long plane = (long long)H * W;
+  const ushort* xb = x + (long long)b * Cin * plane;
+  const ushort* db = dpre + (long long)b * Cout * plane;
+
+  f32x4 acc[NTAP][2] = {};
+  const int kgrp = (lane >> 4) * 8;
+
+  for (int oy = 0; oy < H; ++oy) {
+    __syncthreads();
+    // ---- stage dp rows [64 co] at px [-8, 48): 7 aligned b128 per row
+    // (the kx=0 fragment at kgrp=24 reaches one pixel past the chunk)
+    for (int u = tid; u < 64 * 7; u += 256) {
+      const int co = u / 7, blk = u % 7;
+      const int gco = co0 + co;
+      const int px0 = ux0 - 8 + blk * 8;
+      ushort vals[8] = {};
+      if (gco < Cout) {
+        const ushort* src = db + gco * plane + (long long)oy * W;
+        if (px0 >= 0 && px0 + 8 <= W) {
+          *reinterpret_cast<s16x8*>(vals) =
+              *reinterpret_cast<const s16x8*>(src + px0);
+        } else {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int px = px0 + e;
+            vals[e] = (px >= 0 && px < W) ? src[px] : (ushort)0;
+          }
+        }
+      }
+      *reinterpret_cast<s16x8*>(&dp[co * WGV2_DPW + blk * 8]) =
+          *reinterpret_cast<const s16x8*>(vals);
+    }
+    // ---- stage X rows [32 ci][KS rows] at px [0, 32): 4 b128 per row
+    for (int u = tid; u < 32 * KS * 4; u += 256) {
+      const int ci = u / (KS * 4), ky = (u / 4) % KS, blk = u % 4;
+      const int gci = ci0 + ci;
+      const int iy = oy + ky - PAD;
+      const int px0 = ux0 + blk * 8;
+      ushort vals[8] = {};
+      if (gci < Cin && iy >= 0 && iy < H) {
+        const ushort* src = xb + gci * plane + (long long)iy * W;
+        if (px0 + 8 <= W) {
+          *reinterpret_cast<s16x8*>(vals) =
+              *reinterpret_cast<const s16x8*>(src + px0);
+        } else {
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            vals[e] = (px0 + e < W) ? src[px0 + e] : (ushort)0;
+        }
+      }
+      *reinterpret_cast<s16x8*>(&xs[(ci * KS + ky) * WGV2_XW + blk * 8]) =
+          *reinterpret_cast<const s16x8*>(vals);
+    }
+    __syncthreads();
+
+    // ---- compute: wave w covers co rows [co0+w*16, +16)
+    const ushort* dprow = &dp[(wave * 16 + (lane & 15)) * WGV2_DPW + 8 + kgrp];
+    const s16x8 a_m = *reinterpret_cast<const s16x8*>(dprow - 8);
+    const s16x8 a_0 = *reinterpret_cast<const s16x8*>(dprow);
+    const s16x8 a_p = *reinterpret_cast<const s16x8*>(dprow + 8);
+    s16x8 afrag[NTAP == 1 ? 1 : 3];
+    if (KS == 1) {
+      afrag[0] = a_0;
+    } else {
+      // dp index = u + PAD - kx: kx=0 -> +1, kx=1 -> 0, kx=2 -> -1
+      afrag[0] = __builtin_shufflevector(a_0, a_p, 1, 2, 3, 4, 5, 6, 7, 8);
+      afrag[1] = a_0;
+      afrag[2] = __builtin_shufflevector(a_m, a_0, 7, 8, 9, 10, 11, 12, 13, 14);
+    }
+#pragma unroll
+    for (int ky = 0; ky < KS; ++ky) {
+#pragma unroll
+      for (int nci = 0; nci < 2; ++nci) {
+        const s16x8 bfrag = *reinterpret_cast<const s16x8*>(
+            &xs[((nci * 16 + (lane & 15)) * KS + ky) * WGV2_XW + kgrp]);
+#pragma unroll
+        for (int kx = 0; kx < KS; ++kx)
+          acc[ky * KS + kx][nci] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[kx], bfrag, acc[ky * KS + kx][nci], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- flush: each wave owns distinct co rows -> direct atomics
+#pragma unroll
+  for (int tap = 0; tap < NTAP; ++tap) {
+    const int ky = tap / KS, kx = tap % KS;
+#pragma unroll
+    for (int nci = 0; nci < 2; ++nci) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int co = co0 + wave * 16 + (lane >> 4) * 4 + j;
+        const int ci = ci0 + nci * 16 + (lane & 15);
+        if (co < Cout_p && ci < Cin_p)
+          atomicAdd(&dwp[(((long long)co * Cin_p + ci) * KS + ky) * KS + kx],
+                    acc[tap][nci][j]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Activation backward: dpre = dy * act'(y) elementwise, bf16.
 // ---------------------------------------------------------------------------
 
@@ -560,20 +698,33 @@ at::Tensor conv2d_wgrad_mfma(const at::Tensor& x, const at::Tensor& dpre,
   TORCH_CHECK(Cin_p % 16 == 0 && Cout_p % 16 == 0);
   auto dwp = at::zeros({Cout_p, Cin_p, ks, ks},
                        x.options().dtype(at::kFloat));
-  const int rows_per_blk = 16;
   const int uw = (W + TW - 1) / TW;
-  const long long nslab =
-      (long long)B * ((Ho + rows_per_blk - 1) / rows_per_blk) * uw;
-  dim3 grid((unsigned)nslab, Cin_p / 16, Cout_p / 16);
   auto stream = at::hip::getCurrentHIPStream();
-  DISPATCH_KS_STRIDE((int)ks, (int)stride, [&] {
-    hipLaunchKernelGGL((conv2d_wgrad_mfma_kernel<kKS, kST>),
+  if (stride == 1) {
+    dim3 grid((unsigned)((long long)B * uw), (Cin_p + 31) / 32,
+              (Cout_p + 63) / 64);
+    if (ks == 3)
+      hipLaunchKernelGGL((conv2d_wgrad_s1_kernel<3>), grid, dim3(256), 0,
+                         stream, (const ushort*)x.data_ptr(),
+                         (const ushort*)dpre.data_ptr(), dwp.data_ptr<float>(),
+                         Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw);
+    else
+      hipLaunchKernelGGL((conv2d_wgrad_s1_kernel<1>), grid, dim3(256), 0,
+                         stream, (const ushort*)x.data_ptr(),
+                         (const ushort*)dpre.data_ptr(), dwp.data_ptr<float>(),
+                         Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw);
+  } else {
+    const int rows_per_blk = 16;
+    const long long nslab =
+        (long long)B * ((Ho + rows_per_blk - 1) / rows_per_blk) * uw;
+    dim3 grid((unsigned)nslab, Cin_p / 16, Cout_p / 16);
+    TORCH_CHECK(ks == 3, "stride-2 wgrad supports 3x3 only");
+    hipLaunchKernelGGL((conv2d_wgrad_mfma_kernel<3, 2>),
                        grid, dim3(256), 0, stream,
                        (const ushort*)x.data_ptr(), (const ushort*)dpre.data_ptr(),
                        dwp.data_ptr<float>(), Cin, H, W, Cout, Ho, Wo,
                        (int)Cin_p, (int)Cout_p, rows_per_blk);
-    return 0;
-  });
+  }
   C10_HIP_KERNEL_LAUNCH_CHECK();
   return dwp;
 }
