@@ -408,58 +408,32 @@ void conv2d_wgrad_s1_kernel(
     const ushort* __restrict__ dpre,  // [B, Cout, H, W] (stride 1: Ho=H)
     float* __restrict__ dwp,          // [Cout_p, Cin_p, KS, KS] fp32
     int Cin, int H, int W, int Cout,
-    int Cin_p, int Cout_p, int uw) {
+    int Cin_p, int Cout_p, int uw, int B, int fpb) {
   constexpr int PAD = KS / 2;
   constexpr int NTAP = KS * KS;
   __shared__ ushort dp[64 * WGV2_DPW];        // px domain [-8, 48)
-  __shared__ ushort xs[32 * KS * WGV2_XW];    // px domain [0, 32)
+  __shared__ ushort xs[32 * 3 * WGV2_XW];     // rolling 3-row window
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
 
-  const int b = blockIdx.x / uw;
+  const int b0 = (blockIdx.x / uw) * fpb;
   const int ux0 = (blockIdx.x % uw) * TW;
   const int ci0 = blockIdx.y * 32;
   const int co0 = blockIdx.z * 64;
 
   const long long plane = (long long)H * W;
-  const ushort* xb = x + (long long)b * Cin * plane;
-  const ushort* db = dpre + (long long)b * Cout * plane;
 
   f32x4 acc[NTAP][2] = {};
   const int kgrp = (lane >> 4) * 8;
 
-  for (int oy = 0; oy < H; ++oy) {
-    __syncthreads();
-    // ---- stage dp rows [64 co] at px [-8, 48): 7 aligned b128 per row
-    // (the kx=0 fragment at kgrp=24 reaches one pixel past the chunk)
-    for (int u = tid; u < 64 * 7; u += 256) {
-      const int co = u / 7, blk = u % 7;
-      const int gco = co0 + co;
-      const int px0 = ux0 - 8 + blk * 8;
-      ushort vals[8] = {};
-      if (gco < Cout) {
-        const ushort* src = db + gco * plane + (long long)oy * W;
-        if (px0 >= 0 && px0 + 8 <= W) {
-          *reinterpret_cast<s16x8*>(vals) =
-              *reinterpret_cast<const s16x8*>(src + px0);
-        } else {
-#pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int px = px0 + e;
-            vals[e] = (px >= 0 && px < W) ? src[px] : (ushort)0;
-          }
-        }
-      }
-      *reinterpret_cast<s16x8*>(&dp[co * WGV2_DPW + blk * 8]) =
-          *reinterpret_cast<const s16x8*>(vals);
-    }
-    // ---- stage X rows [32 ci][KS rows] at px [0, 32): 4 b128 per row
-    for (int u = tid; u < 32 * KS * 4; u += 256) {
-      const int ci = u / (KS * 4), ky = (u / 4) % KS, blk = u % 4;
+  // stage one X row (iy) into rolling slot (iy mod 3); zeros when oob
+  auto stage_x_row = [&](const ushort* xb, int iy) {
+    const int slot = ((iy % 3) + 3) % 3;
+    for (int u = tid; u < 32 * 4; u += 256) {
+      const int ci = u >> 2, blk = u & 3;
       const int gci = ci0 + ci;
-      const int iy = oy + ky - PAD;
       const int px0 = ux0 + blk * 8;
       ushort vals[8] = {};
       if (gci < Cin && iy >= 0 && iy < H) {
@@ -473,40 +447,87 @@ void conv2d_wgrad_s1_kernel(
             vals[e] = (px0 + e < W) ? src[px0 + e] : (ushort)0;
         }
       }
-      *reinterpret_cast<s16x8*>(&xs[(ci * KS + ky) * WGV2_XW + blk * 8]) =
+      *reinterpret_cast<s16x8*>(&xs[(ci * 3 + slot) * WGV2_XW + blk * 8]) =
           *reinterpret_cast<const s16x8*>(vals);
     }
-    __syncthreads();
+  };
 
-    // ---- compute: wave w covers co rows [co0+w*16, +16)
-    const ushort* dprow = &dp[(wave * 16 + (lane & 15)) * WGV2_DPW + 8 + kgrp];
-    const s16x8 a_m = *reinterpret_cast<const s16x8*>(dprow - 8);
-    const s16x8 a_0 = *reinterpret_cast<const s16x8*>(dprow);
-    const s16x8 a_p = *reinterpret_cast<const s16x8*>(dprow + 8);
-    s16x8 afrag[NTAP == 1 ? 1 : 3];
-    if (KS == 1) {
-      afrag[0] = a_0;
-    } else {
-      // dp index = u + PAD - kx: kx=0 -> +1, kx=1 -> 0, kx=2 -> -1
-      afrag[0] = __builtin_shufflevector(a_0, a_p, 1, 2, 3, 4, 5, 6, 7, 8);
-      afrag[1] = a_0;
-      afrag[2] = __builtin_shufflevector(a_m, a_0, 7, 8, 9, 10, 11, 12, 13, 14);
+  for (int bf = 0; bf < fpb && b0 + bf < B; ++bf) {
+    const int b = b0 + bf;
+    const ushort* xb = x + (long long)b * Cin * plane;
+    const ushort* db = dpre + (long long)b * Cout * plane;
+
+    // prologue: rows -1 (zeros via oob) and 0 of the window
+    __syncthreads();
+    if (KS == 3) {
+      stage_x_row(xb, -1);
+      stage_x_row(xb, 0);
     }
+
+    for (int oy = 0; oy < H; ++oy) {
+      // ---- stage dp row [64 co] at px [-8, 48): 7 aligned b128 per row
+      // (the kx=0 fragment at kgrp=24 reaches one pixel past the chunk)
+      for (int u = tid; u < 64 * 7; u += 256) {
+        const int co = u / 7, blk = u % 7;
+        const int gco = co0 + co;
+        const int px0 = ux0 - 8 + blk * 8;
+        ushort vals[8] = {};
+        if (gco < Cout) {
+          const ushort* src = db + gco * plane + (long long)oy * W;
+          if (px0 >= 0 && px0 + 8 <= W) {
+            *reinterpret_cast<s16x8*>(vals) =
+                *reinterpret_cast<const s16x8*>(src + px0);
+          } else {
 #pragma unroll
-    for (int ky = 0; ky < KS; ++ky) {
-#pragma unroll
-      for (int nci = 0; nci < 2; ++nci) {
-        const s16x8 bfrag = *reinterpret_cast<const s16x8*>(
-            &xs[((nci * 16 + (lane & 15)) * KS + ky) * WGV2_XW + kgrp]);
-#pragma unroll
-        for (int kx = 0; kx < KS; ++kx)
-          acc[ky * KS + kx][nci] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag[kx], bfrag, acc[ky * KS + kx][nci], 0, 0, 0);
+            for (int e = 0; e < 8; ++e) {
+              const int px = px0 + e;
+              vals[e] = (px >= 0 && px < W) ? src[px] : (ushort)0;
+            }
+          }
+        }
+        *reinterpret_cast<s16x8*>(&dp[co * WGV2_DPW + blk * 8]) =
+            *reinterpret_cast<const s16x8*>(vals);
       }
+      // ---- stage the NEW X row of the sliding window (iy = oy + 1 for
+      // 3x3; the row oy itself for 1x1)
+      stage_x_row(xb, KS == 3 ? oy + 1 : oy);
+      __syncthreads();
+
+      // ---- compute: wave w covers co rows [co0+w*16, +16)
+      const ushort* dprow =
+          &dp[(wave * 16 + (lane & 15)) * WGV2_DPW + 8 + kgrp];
+      const s16x8 a_0 = *reinterpret_cast<const s16x8*>(dprow);
+      s16x8 afrag[NTAP == 1 ? 1 : 3];
+      if (KS == 1) {
+        afrag[0] = a_0;
+      } else {
+        const s16x8 a_m = *reinterpret_cast<const s16x8*>(dprow - 8);
+        const s16x8 a_p = *reinterpret_cast<const s16x8*>(dprow + 8);
+        // dp index = u + PAD - kx: kx=0 -> +1, kx=1 -> 0, kx=2 -> -1
+        afrag[0] = __builtin_shufflevector(a_0, a_p, 1, 2, 3, 4, 5, 6, 7, 8);
+        afrag[1] = a_0;
+        afrag[2] =
+            __builtin_shufflevector(a_m, a_0, 7, 8, 9, 10, 11, 12, 13, 14);
+      }
+#pragma unroll
+      for (int ky = 0; ky < KS; ++ky) {
+        const int slot = KS == 1 ? oy % 3 : (((oy + ky - PAD) % 3) + 3) % 3;
+#pragma unroll
+        for (int nci = 0; nci < 2; ++nci) {
+          const s16x8 bfrag = *reinterpret_cast<const s16x8*>(
+              &xs[((nci * 16 + (lane & 15)) * 3 + slot) * WGV2_XW + kgrp]);
+#pragma unroll
+          for (int kx = 0; kx < KS; ++kx)
+            acc[ky * KS + kx][nci] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[kx], bfrag, acc[ky * KS + kx][nci], 0, 0, 0);
+        }
+      }
+      __syncthreads();
     }
   }
 
   // ---- flush: each wave owns distinct co rows -> direct atomics
+  // (fpb frames accumulated in registers -> B/fpb contributions/address)
 #pragma unroll
   for (int tap = 0; tap < NTAP; ++tap) {
     const int ky = tap / KS, kx = tap % KS;
@@ -701,18 +722,24 @@ at::Tensor conv2d_wgrad_mfma(const at::Tensor& x, const at::Tensor& dpre,
   const int uw = (W + TW - 1) / TW;
   auto stream = at::hip::getCurrentHIPStream();
   if (stride == 1) {
-    dim3 grid((unsigned)((long long)B * uw), (Cin_p + 31) / 32,
-              (Cout_p + 63) / 64);
+    const int ciblks = (Cin_p + 31) / 32, coblks = (Cout_p + 63) / 64;
+    // frames-per-block: accumulate several frames in registers before the
+    // atomic flush (atomic traffic / fpb) while keeping the grid >= ~2048
+    long long nb = (long long)B * uw * ciblks * coblks;
+    int fpb = 1;
+    while (fpb < 16 && fpb * 2 <= B && nb / (fpb * 2) >= 2048) fpb *= 2;
+    const int bblks = (B + fpb - 1) / fpb;
+    dim3 grid((unsigned)((long long)bblks * uw), ciblks, coblks);
     if (ks == 3)
       hipLaunchKernelGGL((conv2d_wgrad_s1_kernel<3>), grid, dim3(256), 0,
                          stream, (const ushort*)x.data_ptr(),
                          (const ushort*)dpre.data_ptr(), dwp.data_ptr<float>(),
-                         Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw);
+                         Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw, B, fpb);
     else
       hipLaunchKernelGGL((conv2d_wgrad_s1_kernel<1>), grid, dim3(256), 0,
                          stream, (const ushort*)x.data_ptr(),
                          (const ushort*)dpre.data_ptr(), dwp.data_ptr<float>(),
-                         Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw);
+                         Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw, B, fpb);
   } else {
     const int rows_per_blk = 16;
     const long long nslab =

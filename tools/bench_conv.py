@@ -51,11 +51,48 @@ def bench(fn, iters, warmup=10):
     return (time.perf_counter() - t0) / iters * 1e3
 
 
+def wgrad_component(args):
+    """Isolated weight-grad: native conv2d_wgrad_mfma vs aten wgrad-only."""
+    from esr_amd.ops.native import require_ext
+    ext = require_ext()
+    dev = "cuda:0"
+    torch.manual_seed(0)
+
+    def ceil(v, m):
+        return (v + m - 1) // m * m
+
+    print(f"{'shape':34s} {'native ms':>10s} {'aten ms':>10s} {'x':>6s}")
+    for label, B, cin, cout, h, w, ks, stride, _ in SHAPES:
+        x = torch.randn(B, cin, h, w, device=dev).to(torch.bfloat16)
+        dy_h = (h + 2 * (ks // 2) - ks) // stride + 1
+        dy_w = (w + 2 * (ks // 2) - ks) // stride + 1
+        dy = torch.randn(B, cout, dy_h, dy_w, device=dev).to(torch.bfloat16)
+        wt = torch.randn(cout, cin, ks, ks, device=dev).to(torch.bfloat16)
+
+        def native():
+            return ext.conv2d_wgrad_mfma(x, dy, ks, stride,
+                                         ceil(cin, 16), ceil(cout, 16))
+
+        def aten():
+            return torch.ops.aten.convolution_backward(
+                dy, x, wt, None, [stride, stride], [ks // 2, ks // 2],
+                [1, 1], False, [0, 0], 1, [False, True, False])[1]
+
+        tn = bench(native, args.iters)
+        tt = bench(aten, args.iters)
+        print(f"{label:34s} {tn:10.3f} {tt:10.3f} {tt / tn:6.2f}")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=50)
     ap.add_argument("--bwd", action="store_true", help="also time fwd+bwd")
+    ap.add_argument("--wgrad", action="store_true",
+                    help="time the wgrad kernel alone vs aten wgrad-only")
     args = ap.parse_args()
+
+    if args.wgrad:
+        return wgrad_component(args)
 
     from esr_amd.ops.conv import ACT_IDS, _NativeConv2dFn
     dev = "cuda:0"
