@@ -24,44 +24,47 @@
 
 namespace {
 
-ESR_INLINE float bilinear_sample(const float* __restrict__ im, int H, int W,
+template <typename T>
+ESR_INLINE float bilinear_sample(const T* __restrict__ im, int H, int W,
                                  float h, float w) {
   // zero-padded bilinear; caller guarantees h > -1 && w > -1 && h < H && w < W
   int h0 = (int)floorf(h);
   int w0 = (int)floorf(w);
   float lh = h - h0, lw = w - w0;
   float hh = 1.f - lh, hw = 1.f - lw;
-  float v00 = (h0 >= 0 && w0 >= 0) ? im[h0 * W + w0] : 0.f;
-  float v01 = (h0 >= 0 && w0 + 1 < W) ? im[h0 * W + w0 + 1] : 0.f;
-  float v10 = (h0 + 1 < H && w0 >= 0) ? im[(h0 + 1) * W + w0] : 0.f;
-  float v11 = (h0 + 1 < H && w0 + 1 < W) ? im[(h0 + 1) * W + w0 + 1] : 0.f;
+  float v00 = (h0 >= 0 && w0 >= 0) ? esr_to_f32(im[h0 * W + w0]) : 0.f;
+  float v01 = (h0 >= 0 && w0 + 1 < W) ? esr_to_f32(im[h0 * W + w0 + 1]) : 0.f;
+  float v10 = (h0 + 1 < H && w0 >= 0) ? esr_to_f32(im[(h0 + 1) * W + w0]) : 0.f;
+  float v11 = (h0 + 1 < H && w0 + 1 < W) ? esr_to_f32(im[(h0 + 1) * W + w0 + 1]) : 0.f;
   return hh * hw * v00 + hh * lw * v01 + lh * hw * v10 + lh * lw * v11;
 }
 
 // gradient of bilinear_sample wrt the sample coordinates
-ESR_INLINE float coord_grad_h(const float* __restrict__ im, int H, int W,
+template <typename T>
+ESR_INLINE float coord_grad_h(const T* __restrict__ im, int H, int W,
                               float h, float w) {
   if (h <= -1 || w <= -1 || h >= H || w >= W) return 0.f;
   int h0 = (int)floorf(h);
   int w0 = (int)floorf(w);
   float lw = w - w0, hw = 1.f - lw;
-  float v00 = (h0 >= 0 && w0 >= 0) ? im[h0 * W + w0] : 0.f;
-  float v01 = (h0 >= 0 && w0 + 1 < W) ? im[h0 * W + w0 + 1] : 0.f;
-  float v10 = (h0 + 1 < H && w0 >= 0) ? im[(h0 + 1) * W + w0] : 0.f;
-  float v11 = (h0 + 1 < H && w0 + 1 < W) ? im[(h0 + 1) * W + w0 + 1] : 0.f;
+  float v00 = (h0 >= 0 && w0 >= 0) ? esr_to_f32(im[h0 * W + w0]) : 0.f;
+  float v01 = (h0 >= 0 && w0 + 1 < W) ? esr_to_f32(im[h0 * W + w0 + 1]) : 0.f;
+  float v10 = (h0 + 1 < H && w0 >= 0) ? esr_to_f32(im[(h0 + 1) * W + w0]) : 0.f;
+  float v11 = (h0 + 1 < H && w0 + 1 < W) ? esr_to_f32(im[(h0 + 1) * W + w0 + 1]) : 0.f;
   return (v10 - v00) * hw + (v11 - v01) * lw;
 }
 
-ESR_INLINE float coord_grad_w(const float* __restrict__ im, int H, int W,
+template <typename T>
+ESR_INLINE float coord_grad_w(const T* __restrict__ im, int H, int W,
                               float h, float w) {
   if (h <= -1 || w <= -1 || h >= H || w >= W) return 0.f;
   int h0 = (int)floorf(h);
   int w0 = (int)floorf(w);
   float lh = h - h0, hh = 1.f - lh;
-  float v00 = (h0 >= 0 && w0 >= 0) ? im[h0 * W + w0] : 0.f;
-  float v01 = (h0 >= 0 && w0 + 1 < W) ? im[h0 * W + w0 + 1] : 0.f;
-  float v10 = (h0 + 1 < H && w0 >= 0) ? im[(h0 + 1) * W + w0] : 0.f;
-  float v11 = (h0 + 1 < H && w0 + 1 < W) ? im[(h0 + 1) * W + w0 + 1] : 0.f;
+  float v00 = (h0 >= 0 && w0 >= 0) ? esr_to_f32(im[h0 * W + w0]) : 0.f;
+  float v01 = (h0 >= 0 && w0 + 1 < W) ? esr_to_f32(im[h0 * W + w0 + 1]) : 0.f;
+  float v10 = (h0 + 1 < H && w0 >= 0) ? esr_to_f32(im[(h0 + 1) * W + w0]) : 0.f;
+  float v11 = (h0 + 1 < H && w0 + 1 < W) ? esr_to_f32(im[(h0 + 1) * W + w0 + 1]) : 0.f;
   return (v01 - v00) * hh + (v11 - v10) * lh;
 }
 
@@ -70,10 +73,11 @@ struct DcnGeom {
 };
 
 // ---------------------------------------------------------------- im2col
+template <typename T>
 __global__ void dcn_im2col_kernel(
-    long long n, const float* __restrict__ im,
-    const float* __restrict__ offset, const float* __restrict__ mask,
-    DcnGeom g, float* __restrict__ cols) {
+    long long n, const T* __restrict__ im,
+    const T* __restrict__ offset, const T* __restrict__ mask,
+    DcnGeom g, T* __restrict__ cols) {
   const int K = g.kh * g.kw;
   const int HoWo = g.Ho * g.Wo;
   const int cpg = g.C / g.dg;  // channels per deformable group
@@ -85,11 +89,11 @@ __global__ void dcn_im2col_kernel(
     const int b = index / ((long long)HoWo * g.C);
     const int grp = c / cpg;
 
-    const float* im_p = im + ((long long)b * g.C + c) * g.H * g.W;
-    const float* off_p = offset +
+    const T* im_p = im + ((long long)b * g.C + c) * g.H * g.W;
+    const T* off_p = offset +
         ((long long)b * g.dg + grp) * 2 * K * HoWo;
-    const float* msk_p = mask + ((long long)b * g.dg + grp) * K * HoWo;
-    float* col_p = cols + (((long long)b * g.C + c) * K * HoWo)
+    const T* msk_p = mask + ((long long)b * g.dg + grp) * K * HoWo;
+    T* col_p = cols + (((long long)b * g.C + c) * K * HoWo)
         + ho * g.Wo + wo;
 
     const int h_in = ho * g.sh - g.ph;
@@ -100,15 +104,15 @@ __global__ void dcn_im2col_kernel(
     for (int i = 0; i < g.kh; ++i) {
       for (int j = 0; j < g.kw; ++j) {
         const int k = i * g.kw + j;
-        const float off_h = off_p[(2 * k) * HoWo + pix];
-        const float off_w = off_p[(2 * k + 1) * HoWo + pix];
-        const float m = msk_p[k * HoWo + pix];
+        const float off_h = esr_to_f32(off_p[(2 * k) * HoWo + pix]);
+        const float off_w = esr_to_f32(off_p[(2 * k + 1) * HoWo + pix]);
+        const float m = esr_to_f32(msk_p[k * HoWo + pix]);
         const float h_im = h_in + i * g.dh + off_h;
         const float w_im = w_in + j * g.dw + off_w;
         float val = 0.f;
         if (h_im > -1 && w_im > -1 && h_im < g.H && w_im < g.W)
           val = bilinear_sample(im_p, g.H, g.W, h_im, w_im);
-        col_p[(long long)k * HoWo] = val * m;
+        col_p[(long long)k * HoWo] = esr_from_f32<T>(val * m);
       }
     }
   }
@@ -126,10 +130,11 @@ constexpr int C2I_TILE = 16;
 constexpr int C2I_HALO = 4;
 constexpr int C2I_EDGE = C2I_TILE + 2 * C2I_HALO;   // 24
 
+template <typename T>
 __global__ __launch_bounds__(256)
 void dcn_col2im_tiled_kernel(
-    const float* __restrict__ col_grad, const float* __restrict__ offset,
-    const float* __restrict__ mask, DcnGeom g, float* __restrict__ grad_im) {
+    const T* __restrict__ col_grad, const T* __restrict__ offset,
+    const T* __restrict__ mask, DcnGeom g, float* __restrict__ grad_im) {
   const int K = g.kh * g.kw;            // 9
   const int HoWo = g.Ho * g.Wo;
   const int cpg = g.C / g.dg;
@@ -155,19 +160,19 @@ void dcn_col2im_tiled_kernel(
 
   if (ho < g.Ho && wo < g.Wo) {
     const int pix = ho * g.Wo + wo;
-    const float* off_p = offset + ((long long)b * g.dg + grp) * 2 * K * HoWo;
-    const float* msk_p = mask + ((long long)b * g.dg + grp) * K * HoWo;
-    const float* cg_p = col_grad + (((long long)b * g.C + c) * K) * HoWo + pix;
+    const T* off_p = offset + ((long long)b * g.dg + grp) * 2 * K * HoWo;
+    const T* msk_p = mask + ((long long)b * g.dg + grp) * K * HoWo;
+    const T* cg_p = col_grad + (((long long)b * g.C + c) * K) * HoWo + pix;
     #pragma unroll
     for (int k = 0; k < 9; ++k) {
       const int i = k / 3, j = k % 3;
-      const float off_h = off_p[(2 * k) * HoWo + pix];
-      const float off_w = off_p[(2 * k + 1) * HoWo + pix];
-      const float m = msk_p[k * HoWo + pix];
+      const float off_h = esr_to_f32(off_p[(2 * k) * HoWo + pix]);
+      const float off_w = esr_to_f32(off_p[(2 * k + 1) * HoWo + pix]);
+      const float m = esr_to_f32(msk_p[k * HoWo + pix]);
       const float h_im = ho - 1 + i + off_h;
       const float w_im = wo - 1 + j + off_w;
       if (h_im <= -1 || w_im <= -1 || h_im >= g.H || w_im >= g.W) continue;
-      const float gval = cg_p[(long long)k * HoWo] * m;
+      const float gval = esr_to_f32(cg_p[(long long)k * HoWo]) * m;
       const int h0 = (int)floorf(h_im);
       const int w0 = (int)floorf(w_im);
       const float lh = h_im - h0, lw = w_im - w0;
@@ -203,9 +208,10 @@ void dcn_col2im_tiled_kernel(
 // ------------------------------------------------------------- col2im
 // grad wrt input: distribute each column grad over its <=4 integer
 // neighbours with bilinear weights; atomicAdd into grad_im (device scope).
+template <typename T>
 __global__ void dcn_col2im_kernel(
-    long long n, const float* __restrict__ col_grad,
-    const float* __restrict__ offset, const float* __restrict__ mask,
+    long long n, const T* __restrict__ col_grad,
+    const T* __restrict__ offset, const T* __restrict__ mask,
     DcnGeom g, float* __restrict__ grad_im) {
   const int K = g.kh * g.kw;
   const int HoWo = g.Ho * g.Wo;
@@ -221,17 +227,19 @@ __global__ void dcn_col2im_kernel(
     const int i = k / g.kw, j = k % g.kw;
     const int pix = ho * g.Wo + wo;
 
-    const float* off_p = offset + ((long long)b * g.dg + grp) * 2 * K * HoWo;
-    const float off_h = off_p[(2 * k) * HoWo + pix];
-    const float off_w = off_p[(2 * k + 1) * HoWo + pix];
-    const float m = mask[(((long long)b * g.dg + grp) * K + k) * HoWo + pix];
+    const T* off_p = offset + ((long long)b * g.dg + grp) * 2 * K * HoWo;
+    const float off_h = esr_to_f32(off_p[(2 * k) * HoWo + pix]);
+    const float off_w = esr_to_f32(off_p[(2 * k + 1) * HoWo + pix]);
+    const float m =
+        esr_to_f32(mask[(((long long)b * g.dg + grp) * K + k) * HoWo + pix]);
 
     const float h_im = ho * g.sh - g.ph + i * g.dh + off_h;
     const float w_im = wo * g.sw - g.pw + j * g.dw + off_w;
     if (h_im <= -1 || w_im <= -1 || h_im >= g.H || w_im >= g.W) continue;
 
     const float gval =
-        col_grad[(((long long)b * g.C + c) * K + k) * HoWo + pix] * m;
+        esr_to_f32(col_grad[(((long long)b * g.C + c) * K + k) * HoWo + pix])
+        * m;
     const int h0 = (int)floorf(h_im);
     const int w0 = (int)floorf(w_im);
     const float lh = h_im - h0, lw = w_im - w0;
@@ -250,11 +258,12 @@ __global__ void dcn_col2im_kernel(
 // -------------------------------------------------------- col2im_coord
 // grad wrt offsets and mask: per (b, grp, k, ho, wo) reduce over the
 // group's channels; plain stores (no atomics).
+template <typename T>
 __global__ void dcn_col2im_coord_kernel(
-    long long n, const float* __restrict__ col_grad,
-    const float* __restrict__ im, const float* __restrict__ offset,
-    const float* __restrict__ mask, DcnGeom g,
-    float* __restrict__ grad_offset, float* __restrict__ grad_mask) {
+    long long n, const T* __restrict__ col_grad,
+    const T* __restrict__ im, const T* __restrict__ offset,
+    const T* __restrict__ mask, DcnGeom g,
+    T* __restrict__ grad_offset, T* __restrict__ grad_mask) {
   const int K = g.kh * g.kw;
   const int HoWo = g.Ho * g.Wo;
   const int cpg = g.C / g.dg;
@@ -268,10 +277,11 @@ __global__ void dcn_col2im_coord_kernel(
     const int i = k / g.kw, j = k % g.kw;
     const int pix = ho * g.Wo + wo;
 
-    const float* off_p = offset + ((long long)b * g.dg + grp) * 2 * K * HoWo;
-    const float off_h = off_p[(2 * k) * HoWo + pix];
-    const float off_w = off_p[(2 * k + 1) * HoWo + pix];
-    const float m = mask[(((long long)b * g.dg + grp) * K + k) * HoWo + pix];
+    const T* off_p = offset + ((long long)b * g.dg + grp) * 2 * K * HoWo;
+    const float off_h = esr_to_f32(off_p[(2 * k) * HoWo + pix]);
+    const float off_w = esr_to_f32(off_p[(2 * k + 1) * HoWo + pix]);
+    const float m =
+        esr_to_f32(mask[(((long long)b * g.dg + grp) * K + k) * HoWo + pix]);
 
     const float h_im = ho * g.sh - g.ph + i * g.dh + off_h;
     const float w_im = wo * g.sw - g.pw + j * g.dw + off_w;
@@ -281,18 +291,19 @@ __global__ void dcn_col2im_coord_kernel(
     for (int cc = 0; cc < cpg; ++cc) {
       const int c = grp * cpg + cc;
       const float cg =
-          col_grad[(((long long)b * g.C + c) * K + k) * HoWo + pix];
-      const float* im_p = im + ((long long)b * g.C + c) * g.H * g.W;
+          esr_to_f32(col_grad[(((long long)b * g.C + c) * K + k) * HoWo + pix]);
+      const T* im_p = im + ((long long)b * g.C + c) * g.H * g.W;
       if (in_range) {
         gh += cg * m * coord_grad_h(im_p, g.H, g.W, h_im, w_im);
         gw += cg * m * coord_grad_w(im_p, g.H, g.W, h_im, w_im);
         gm += cg * bilinear_sample(im_p, g.H, g.W, h_im, w_im);
       }
     }
-    float* goff = grad_offset + ((long long)b * g.dg + grp) * 2 * K * HoWo;
-    goff[(2 * k) * HoWo + pix] = gh;
-    goff[(2 * k + 1) * HoWo + pix] = gw;
-    grad_mask[(((long long)b * g.dg + grp) * K + k) * HoWo + pix] = gm;
+    T* goff = grad_offset + ((long long)b * g.dg + grp) * 2 * K * HoWo;
+    goff[(2 * k) * HoWo + pix] = esr_from_f32<T>(gh);
+    goff[(2 * k + 1) * HoWo + pix] = esr_from_f32<T>(gw);
+    grad_mask[(((long long)b * g.dg + grp) * K + k) * HoWo + pix] =
+        esr_from_f32<T>(gm);
   }
 }
 
@@ -308,16 +319,34 @@ DcnGeom make_geom(const at::Tensor& input, const at::Tensor& weight,
   return g;
 }
 
+// fp32 or bf16 (fp32 compute inside either way)
+#define DISPATCH_DCN_DTYPE(TEN, ...)                                   \
+  [&] {                                                                \
+    if ((TEN).scalar_type() == at::kFloat) {                           \
+      using scalar_t = float;                                          \
+      return __VA_ARGS__();                                            \
+    }                                                                  \
+    TORCH_CHECK((TEN).scalar_type() == at::kBFloat16,                  \
+                "deform_conv2d: fp32 or bf16 required");               \
+    using scalar_t = __hip_bfloat16;                                   \
+    return __VA_ARGS__();                                              \
+  }()
+
 at::Tensor dcn_im2col(const at::Tensor& input, const at::Tensor& offset,
                       const at::Tensor& mask, const DcnGeom& g) {
   auto cols = at::empty({g.B, g.C * g.kh * g.kw, g.Ho * g.Wo},
                         input.options());
   long long n = (long long)g.B * g.C * g.Ho * g.Wo;
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(dcn_im2col_kernel, dim3(esr_grid(n)), dim3(ESR_BLOCK), 0,
-                     stream, n, input.data_ptr<float>(),
-                     offset.data_ptr<float>(), mask.data_ptr<float>(), g,
-                     cols.data_ptr<float>());
+  DISPATCH_DCN_DTYPE(input, [&] {
+    hipLaunchKernelGGL((dcn_im2col_kernel<scalar_t>), dim3(esr_grid(n)),
+                       dim3(ESR_BLOCK), 0, stream, n,
+                       (const scalar_t*)input.data_ptr(),
+                       (const scalar_t*)offset.data_ptr(),
+                       (const scalar_t*)mask.data_ptr(), g,
+                       (scalar_t*)cols.data_ptr());
+    return 0;
+  });
   return cols;
 }
 
@@ -336,8 +365,9 @@ at::Tensor deform_conv2d_forward(
     const at::Tensor& weight, const c10::optional<at::Tensor>& bias,
     int64_t sh, int64_t sw, int64_t ph, int64_t pw, int64_t dh, int64_t dw,
     int64_t dg) {
-  TORCH_CHECK(input.is_cuda() && input.scalar_type() == at::kFloat,
-              "deform_conv2d: fp32 CUDA tensors required");
+  TORCH_CHECK(input.is_cuda() && (input.scalar_type() == at::kFloat ||
+                                  input.scalar_type() == at::kBFloat16),
+              "deform_conv2d: fp32/bf16 CUDA tensors required");
   TORCH_CHECK(input.is_contiguous() && offset.is_contiguous() &&
               mask.is_contiguous() && weight.is_contiguous());
   auto g = make_geom(input, weight, sh, sw, ph, pw, dh, dw, dg);
@@ -355,7 +385,7 @@ at::Tensor deform_conv2d_forward(
     const char* e = getenv("ESR_DCN_FUSED");
     return e != nullptr && e[0] == '1';
   }();
-  if (use_fused &&
+  if (use_fused && input.scalar_type() == at::kFloat &&
       dcn_fused_applicable(input, weight, sh, sw, ph, pw, dh, dw, dg))
     return deform_conv2d_forward_fused(input, offset, mask, weight, bias, dg);
 
@@ -394,36 +424,49 @@ std::vector<at::Tensor> deform_conv2d_backward(
     const char* e = getenv("ESR_DCN_TILED");
     return e == nullptr || e[0] != '0';
   }();
-  auto grad_input = at::zeros_like(input);
-  if (use_tiled && g.kh == 3 && g.kw == 3 && g.sh == 1 && g.sw == 1 &&
-      g.ph == 1 && g.pw == 1 && g.dh == 1 && g.dw == 1) {
-    const int tiles = ((g.Ho + C2I_TILE - 1) / C2I_TILE) *
-                      ((g.Wo + C2I_TILE - 1) / C2I_TILE);
-    hipLaunchKernelGGL(dcn_col2im_tiled_kernel,
-                       dim3(tiles, g.C, g.B), dim3(256), 0, stream,
-                       col_grad.data_ptr<float>(), offset.data_ptr<float>(),
-                       mask.data_ptr<float>(), g,
-                       grad_input.data_ptr<float>());
-  } else {
-    long long n = (long long)g.B * g.C * K * g.Ho * g.Wo;
-    hipLaunchKernelGGL(dcn_col2im_kernel, dim3(esr_grid(n)), dim3(ESR_BLOCK),
-                       0, stream, n, col_grad.data_ptr<float>(),
-                       offset.data_ptr<float>(), mask.data_ptr<float>(), g,
-                       grad_input.data_ptr<float>());
-  }
+  // atomics accumulate in fp32 regardless of the op dtype
+  auto grad_input_f = at::zeros(input.sizes(), input.options()
+                                                   .dtype(at::kFloat));
+  DISPATCH_DCN_DTYPE(input, [&] {
+    if (use_tiled && g.kh == 3 && g.kw == 3 && g.sh == 1 && g.sw == 1 &&
+        g.ph == 1 && g.pw == 1 && g.dh == 1 && g.dw == 1) {
+      const int tiles = ((g.Ho + C2I_TILE - 1) / C2I_TILE) *
+                        ((g.Wo + C2I_TILE - 1) / C2I_TILE);
+      hipLaunchKernelGGL((dcn_col2im_tiled_kernel<scalar_t>),
+                         dim3(tiles, g.C, g.B), dim3(256), 0, stream,
+                         (const scalar_t*)col_grad.data_ptr(),
+                         (const scalar_t*)offset.data_ptr(),
+                         (const scalar_t*)mask.data_ptr(), g,
+                         grad_input_f.data_ptr<float>());
+    } else {
+      long long n = (long long)g.B * g.C * K * g.Ho * g.Wo;
+      hipLaunchKernelGGL((dcn_col2im_kernel<scalar_t>), dim3(esr_grid(n)),
+                         dim3(ESR_BLOCK), 0, stream, n,
+                         (const scalar_t*)col_grad.data_ptr(),
+                         (const scalar_t*)offset.data_ptr(),
+                         (const scalar_t*)mask.data_ptr(), g,
+                         grad_input_f.data_ptr<float>());
+    }
+    return 0;
+  });
+  auto grad_input = input.scalar_type() == at::kFloat
+      ? grad_input_f : grad_input_f.to(input.scalar_type());
 
   // grad offset + mask
   auto grad_offset = at::empty_like(offset);
   auto grad_mask = at::empty_like(mask);
-  {
+  DISPATCH_DCN_DTYPE(input, [&] {
     long long n = (long long)g.B * g.dg * K * g.Ho * g.Wo;
-    hipLaunchKernelGGL(dcn_col2im_coord_kernel, dim3(esr_grid(n)),
-                       dim3(ESR_BLOCK), 0, stream, n,
-                       col_grad.data_ptr<float>(), input.data_ptr<float>(),
-                       offset.data_ptr<float>(), mask.data_ptr<float>(), g,
-                       grad_offset.data_ptr<float>(),
-                       grad_mask.data_ptr<float>());
-  }
+    hipLaunchKernelGGL((dcn_col2im_coord_kernel<scalar_t>),
+                       dim3(esr_grid(n)), dim3(ESR_BLOCK), 0, stream, n,
+                       (const scalar_t*)col_grad.data_ptr(),
+                       (const scalar_t*)input.data_ptr(),
+                       (const scalar_t*)offset.data_ptr(),
+                       (const scalar_t*)mask.data_ptr(), g,
+                       (scalar_t*)grad_offset.data_ptr(),
+                       (scalar_t*)grad_mask.data_ptr());
+    return 0;
+  });
 
   // grad weight / bias via batched GEMM (the reference loops per sample,
   // ESR:models/DCNv2/src/cuda/dcn_v2_cuda.cu:150)
