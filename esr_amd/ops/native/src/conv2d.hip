@@ -408,7 +408,9 @@ void conv2d_wgrad_s1_kernel(
     const ushort* __restrict__ dpre,  // [B, Cout, H, W] (stride 1: Ho=H)
     float* __restrict__ dwp,          // [Cout_p, Cin_p, KS, KS] fp32
     int Cin, int H, int W, int Cout,
-    int Cin_p, int Cout_p, int uw, int B, int fpb) {
+    int Cin_p, int Cout_p, int uw, int B, int fpb, int abl) {
+  // abl (ablation, tools/bench_conv.py --wgrad): 0 = full, 1 = skip the
+  // atomic flush, 2 = skip MFMA + flush (staging only)
   constexpr int PAD = KS / 2;
   constexpr int NTAP = KS * KS;
   __shared__ ushort dp[64 * WGV2_DPW];        // px domain [-8, 48)
@@ -509,6 +511,7 @@ void conv2d_wgrad_s1_kernel(
         afrag[2] =
             __builtin_shufflevector(a_m, a_0, 7, 8, 9, 10, 11, 12, 13, 14);
       }
+      if (abl == 2) { __syncthreads(); continue; }
 #pragma unroll
       for (int ky = 0; ky < KS; ++ky) {
         const int slot = KS == 1 ? oy % 3 : (((oy + ky - PAD) % 3) + 3) % 3;
@@ -528,6 +531,7 @@ void conv2d_wgrad_s1_kernel(
 
   // ---- flush: each wave owns distinct co rows -> direct atomics
   // (fpb frames accumulated in registers -> B/fpb contributions/address)
+  if (abl == 1) return;
 #pragma unroll
   for (int tap = 0; tap < NTAP; ++tap) {
     const int ky = tap / KS, kx = tap % KS;
@@ -728,18 +732,24 @@ at::Tensor conv2d_wgrad_mfma(const at::Tensor& x, const at::Tensor& dpre,
     long long nb = (long long)B * uw * ciblks * coblks;
     int fpb = 1;
     while (fpb < 16 && fpb * 2 <= B && nb / (fpb * 2) >= 2048) fpb *= 2;
+    static const int abl = [] {
+      const char* e = getenv("ESR_WGRAD_ABL");
+      return e ? atoi(e) : 0;
+    }();
     const int bblks = (B + fpb - 1) / fpb;
     dim3 grid((unsigned)((long long)bblks * uw), ciblks, coblks);
     if (ks == 3)
       hipLaunchKernelGGL((conv2d_wgrad_s1_kernel<3>), grid, dim3(256), 0,
                          stream, (const ushort*)x.data_ptr(),
                          (const ushort*)dpre.data_ptr(), dwp.data_ptr<float>(),
-                         Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw, B, fpb);
+                         Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw, B,
+                         fpb, abl);
     else
       hipLaunchKernelGGL((conv2d_wgrad_s1_kernel<1>), grid, dim3(256), 0,
                          stream, (const ushort*)x.data_ptr(),
                          (const ushort*)dpre.data_ptr(), dwp.data_ptr<float>(),
-                         Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw, B, fpb);
+                         Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw, B,
+                         fpb, abl);
   } else {
     const int rows_per_blk = 16;
     const long long nslab =
