@@ -124,10 +124,11 @@ class _NativeConv2dFn(torch.autograd.Function):
             dx = ext.conv2d_dgrad_s2(dpre, w.contiguous(),
                                      x.shape[2], x.shape[3])
 
-        # weight grad: MFMA split-K into padded fp32, then slice
+        # weight grad: MFMA split-K into padded fp32 (transposed
+        # [Cin_p, Cout_p] scratch for line-parallel atomics), then slice
         dwp = ext.conv2d_wgrad_mfma(x, dpre, ks, stride,
                                     _ceil(cin, 16), _ceil(cout, 16))
-        dw = dwp[:cout, :cin].to(w.dtype)
+        dw = dwp.permute(1, 0, 2, 3)[:cout, :cin].to(w.dtype)
         db = dpre.sum(dim=(0, 2, 3), dtype=torch.float32).to(w.dtype) \
             if ctx.has_bias else None
         return dx, dw, db, None, None

@@ -378,8 +378,10 @@ void conv2d_wgrad_mfma_kernel(
       atomicAdd(&red[((lane >> 4) * 4 + j) * 16 + (lane & 15)], acc[tap][j]);
     __syncthreads();
     if (tid < 256) {
+      // consecutive threads vary ci -> distinct lines in the transposed
+      // [Cin_p][Cout_p] scratch
       const int co = tid >> 4, ci = tid & 15;
-      atomicAdd(&dwp[(((long long)(co0 + co) * Cin_p + ci0 + ci) * KS + ky)
+      atomicAdd(&dwp[(((long long)(ci0 + ci) * Cout_p + co0 + co) * KS + ky)
                      * KS + kx], red[tid]);
     }
     __syncthreads();
@@ -541,8 +543,12 @@ void conv2d_wgrad_s1_kernel(
       for (int j = 0; j < 4; ++j) {
         const int co = co0 + wave * 16 + (lane >> 4) * 4 + j;
         const int ci = ci0 + nci * 16 + (lane & 15);
+        // dwp is TRANSPOSED [Cin_p][Cout_p][ks][ks]: lane&15 varies ci ->
+        // each atomic instruction's 16 lanes hit 16 distinct cachelines
+        // (same-line atomics serialized in L2 were 70-87% of this kernel,
+        // measured via ESR_WGRAD_ABL)
         if (co < Cout_p && ci < Cin_p)
-          atomicAdd(&dwp[(((long long)co * Cin_p + ci) * KS + ky) * KS + kx],
+          atomicAdd(&dwp[(((long long)ci * Cout_p + co) * KS + ky) * KS + kx],
                     acc[tap][nci][j]);
       }
     }
@@ -721,7 +727,9 @@ at::Tensor conv2d_wgrad_mfma(const at::Tensor& x, const at::Tensor& dpre,
   const int B = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
   const int Cout = dpre.size(1), Ho = dpre.size(2), Wo = dpre.size(3);
   TORCH_CHECK(Cin_p % 16 == 0 && Cout_p % 16 == 0);
-  auto dwp = at::zeros({Cout_p, Cin_p, ks, ks},
+  // TRANSPOSED scratch [Cin_p, Cout_p, ks, ks] (see flush comments);
+  // the python layer permutes back when slicing
+  auto dwp = at::zeros({Cin_p, Cout_p, ks, ks},
                        x.options().dtype(at::kFloat));
   const int uw = (W + TW - 1) / TW;
   auto stream = at::hip::getCurrentHIPStream();
@@ -731,7 +739,7 @@ at::Tensor conv2d_wgrad_mfma(const at::Tensor& x, const at::Tensor& dpre,
     // atomic flush (atomic traffic / fpb) while keeping the grid >= ~2048
     long long nb = (long long)B * uw * ciblks * coblks;
     int fpb = 1;
-    while (fpb < 16 && fpb * 2 <= B && nb / (fpb * 2) >= 2048) fpb *= 2;
+    while (fpb < 32 && fpb * 2 <= B && nb / (fpb * 2) >= 1024) fpb *= 2;
     static const int abl = [] {
       const char* e = getenv("ESR_WGRAD_ABL");
       return e ? atoi(e) : 0;
