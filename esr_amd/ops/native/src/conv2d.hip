@@ -531,27 +531,44 @@ void conv2d_wgrad_s1_kernel(
     }
   }
 
-  // ---- flush: each wave owns distinct co rows -> direct atomics
-  // (fpb frames accumulated in registers -> B/fpb contributions/address)
+  // ---- flush: plain coalesced stores of this block's partial tile into
+  // scratch [g][32ci][64co][NTAP]; a second kernel reduces over g.  No
+  // atomics at all (the atomic flood was 70-87% of this kernel, measured
+  // via ESR_WGRAD_ABL) and the reduction order is DETERMINISTIC.
   if (abl == 1) return;
+  float* part = dwp + ((((long long)blockIdx.x * gridDim.y + blockIdx.y)
+                        * gridDim.z + blockIdx.z) * (32 * 64 * NTAP));
 #pragma unroll
   for (int tap = 0; tap < NTAP; ++tap) {
-    const int ky = tap / KS, kx = tap % KS;
 #pragma unroll
     for (int nci = 0; nci < 2; ++nci) {
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        const int co = co0 + wave * 16 + (lane >> 4) * 4 + j;
-        const int ci = ci0 + nci * 16 + (lane & 15);
-        // dwp is TRANSPOSED [Cin_p][Cout_p][ks][ks]: lane&15 varies ci ->
-        // each atomic instruction's 16 lanes hit 16 distinct cachelines
-        // (same-line atomics serialized in L2 were 70-87% of this kernel,
-        // measured via ESR_WGRAD_ABL)
-        if (co < Cout_p && ci < Cin_p)
-          atomicAdd(&dwp[(((long long)ci * Cout_p + co) * KS + ky) * KS + kx],
-                    acc[tap][nci][j]);
+        const int co = wave * 16 + (lane >> 4) * 4 + j;
+        const int ci = nci * 16 + (lane & 15);
+        part[((long long)ci * 64 + co) * NTAP + tap] = acc[tap][nci][j];
       }
     }
+  }
+}
+
+// reduce the per-block partials: dwp[ci][co][tap] = sum_g part[g][...]
+template <int NTAP>
+__global__ void conv2d_wgrad_reduce_kernel(
+    long long n, const float* __restrict__ part, int ng, int ciblks,
+    int coblks, int Cin_p, int Cout_p, float* __restrict__ dwp) {
+  ESR_KERNEL_LOOP(i, n) {  // i -> (ci, co, tap) over the PADDED dW
+    const int tap = (int)(i % NTAP);
+    const int co = (int)((i / NTAP) % Cout_p);
+    const int ci = (int)(i / ((long long)NTAP * Cout_p));
+    const int cib = ci / 32, cob = co / 64;
+    const long long base =
+        ((long long)cib * coblks + cob) * (32 * 64 * NTAP)
+        + (((long long)(ci % 32) * 64 + (co % 64)) * NTAP + tap);
+    const long long gstride = (long long)ciblks * coblks * (32 * 64 * NTAP);
+    float v = 0.f;
+    for (int g = 0; g < ng; ++g) v += part[g * gstride + base];
+    dwp[i] = v;
   }
 }
 
@@ -727,8 +744,8 @@ at::Tensor conv2d_wgrad_mfma(const at::Tensor& x, const at::Tensor& dpre,
   const int B = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
   const int Cout = dpre.size(1), Ho = dpre.size(2), Wo = dpre.size(3);
   TORCH_CHECK(Cin_p % 16 == 0 && Cout_p % 16 == 0);
-  // TRANSPOSED scratch [Cin_p, Cout_p, ks, ks] (see flush comments);
-  // the python layer permutes back when slicing
+  // TRANSPOSED result [Cin_p, Cout_p, ks, ks]; the python layer permutes
+  // back when slicing
   auto dwp = at::zeros({Cin_p, Cout_p, ks, ks},
                        x.options().dtype(at::kFloat));
   const int uw = (W + TW - 1) / TW;
@@ -745,19 +762,36 @@ at::Tensor conv2d_wgrad_mfma(const at::Tensor& x, const at::Tensor& dpre,
       return e ? atoi(e) : 0;
     }();
     const int bblks = (B + fpb - 1) / fpb;
-    dim3 grid((unsigned)((long long)bblks * uw), ciblks, coblks);
+    const int ng = bblks * uw;
+    const int ntap = (int)(ks * ks);
+    dim3 grid((unsigned)ng, ciblks, coblks);
+    auto part = at::empty({(long long)ng * ciblks * coblks * 32 * 64 * ntap},
+                          x.options().dtype(at::kFloat));
     if (ks == 3)
       hipLaunchKernelGGL((conv2d_wgrad_s1_kernel<3>), grid, dim3(256), 0,
                          stream, (const ushort*)x.data_ptr(),
-                         (const ushort*)dpre.data_ptr(), dwp.data_ptr<float>(),
+                         (const ushort*)dpre.data_ptr(),
+                         part.data_ptr<float>(),
                          Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw, B,
                          fpb, abl);
     else
       hipLaunchKernelGGL((conv2d_wgrad_s1_kernel<1>), grid, dim3(256), 0,
                          stream, (const ushort*)x.data_ptr(),
-                         (const ushort*)dpre.data_ptr(), dwp.data_ptr<float>(),
+                         (const ushort*)dpre.data_ptr(),
+                         part.data_ptr<float>(),
                          Cin, H, W, Cout, (int)Cin_p, (int)Cout_p, uw, B,
                          fpb, abl);
+    const long long nred = (long long)Cin_p * Cout_p * ntap;
+    if (ks == 3)
+      hipLaunchKernelGGL((conv2d_wgrad_reduce_kernel<9>),
+                         dim3(esr_grid(nred)), dim3(ESR_BLOCK), 0, stream,
+                         nred, part.data_ptr<float>(), ng, ciblks, coblks,
+                         (int)Cin_p, (int)Cout_p, dwp.data_ptr<float>());
+    else
+      hipLaunchKernelGGL((conv2d_wgrad_reduce_kernel<1>),
+                         dim3(esr_grid(nred)), dim3(ESR_BLOCK), 0, stream,
+                         nred, part.data_ptr<float>(), ng, ciblks, coblks,
+                         (int)Cin_p, (int)Cout_p, dwp.data_ptr<float>());
   } else {
     const int rows_per_blk = 16;
     const long long nslab =
