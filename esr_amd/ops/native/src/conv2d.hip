@@ -408,15 +408,16 @@ __global__ __launch_bounds__(256)
 void conv2d_wgrad_s1_kernel(
     const ushort* __restrict__ x,     // [B, Cin, H, W]
     const ushort* __restrict__ dpre,  // [B, Cout, H, W] (stride 1: Ho=H)
-    float* __restrict__ dwp,          // [Cout_p, Cin_p, KS, KS] fp32
+    float* __restrict__ dwp,          // partial tiles [g][32ci][64co][NTAP]
     int Cin, int H, int W, int Cout,
     int Cin_p, int Cout_p, int uw, int B, int fpb, int abl) {
   // abl (ablation, tools/bench_conv.py --wgrad): 0 = full, 1 = skip the
-  // atomic flush, 2 = skip MFMA + flush (staging only)
+  // flush, 2 = skip MFMA (staging only; flush still runs)
   constexpr int PAD = KS / 2;
   constexpr int NTAP = KS * KS;
-  __shared__ ushort dp[64 * WGV2_DPW];        // px domain [-8, 48)
-  __shared__ ushort xs[32 * 3 * WGV2_XW];     // rolling 3-row window
+  constexpr int RB = 4;                       // output rows per barrier pair
+  __shared__ ushort dp[RB * 64 * WGV2_DPW];   // px domain [-8, 48) per row
+  __shared__ ushort xs[32 * (RB + 2) * WGV2_XW];  // rolling 6-row window
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -431,10 +432,11 @@ void conv2d_wgrad_s1_kernel(
 
   f32x4 acc[NTAP][2] = {};
   const int kgrp = (lane >> 4) * 8;
+  constexpr int NSLOT = RB + 2;
 
-  // stage one X row (iy) into rolling slot (iy mod 3); zeros when oob
+  // stage one X row (iy) into rolling slot (iy mod NSLOT); zeros when oob
   auto stage_x_row = [&](const ushort* xb, int iy) {
-    const int slot = ((iy % 3) + 3) % 3;
+    const int slot = ((iy % NSLOT) + NSLOT) % NSLOT;
     for (int u = tid; u < 32 * 4; u += 256) {
       const int ci = u >> 2, blk = u & 3;
       const int gci = ci0 + ci;
@@ -451,7 +453,7 @@ void conv2d_wgrad_s1_kernel(
             vals[e] = (px0 + e < W) ? src[px0 + e] : (ushort)0;
         }
       }
-      *reinterpret_cast<s16x8*>(&xs[(ci * 3 + slot) * WGV2_XW + blk * 8]) =
+      *reinterpret_cast<s16x8*>(&xs[(ci * NSLOT + slot) * WGV2_XW + blk * 8]) =
           *reinterpret_cast<const s16x8*>(vals);
     }
   };
@@ -461,22 +463,24 @@ void conv2d_wgrad_s1_kernel(
     const ushort* xb = x + (long long)b * Cin * plane;
     const ushort* db = dpre + (long long)b * Cout * plane;
 
-    // prologue: rows -1 (zeros via oob) and 0 of the window
+    // prologue: window rows -1, 0 (zeros via oob for -1)
     __syncthreads();
     if (KS == 3) {
       stage_x_row(xb, -1);
       stage_x_row(xb, 0);
     }
 
-    for (int oy = 0; oy < H; ++oy) {
-      // ---- stage dp row [64 co] at px [-8, 48): 7 aligned b128 per row
+    for (int oy0 = 0; oy0 < H; oy0 += RB) {
+      // ---- stage RB dp rows [64 co] at px [-8, 48): 7 b128 per row
       // (the kx=0 fragment at kgrp=24 reaches one pixel past the chunk)
-      for (int u = tid; u < 64 * 7; u += 256) {
-        const int co = u / 7, blk = u % 7;
+      for (int u = tid; u < RB * 64 * 7; u += 256) {
+        const int r = u / (64 * 7);
+        const int co = (u / 7) % 64, blk = u % 7;
         const int gco = co0 + co;
+        const int oy = oy0 + r;
         const int px0 = ux0 - 8 + blk * 8;
         ushort vals[8] = {};
-        if (gco < Cout) {
+        if (gco < Cout && oy < H) {
           const ushort* src = db + gco * plane + (long long)oy * W;
           if (px0 >= 0 && px0 + 8 <= W) {
             *reinterpret_cast<s16x8*>(vals) =
@@ -489,42 +493,52 @@ void conv2d_wgrad_s1_kernel(
             }
           }
         }
-        *reinterpret_cast<s16x8*>(&dp[co * WGV2_DPW + blk * 8]) =
+        *reinterpret_cast<s16x8*>(&dp[(r * 64 + co) * WGV2_DPW + blk * 8]) =
             *reinterpret_cast<const s16x8*>(vals);
       }
-      // ---- stage the NEW X row of the sliding window (iy = oy + 1 for
-      // 3x3; the row oy itself for 1x1)
-      stage_x_row(xb, KS == 3 ? oy + 1 : oy);
+      // ---- stage the RB new X rows of the sliding window
+#pragma unroll
+      for (int r = 0; r < RB; ++r)
+        stage_x_row(xb, KS == 3 ? oy0 + r + 1 : oy0 + r);
       __syncthreads();
 
-      // ---- compute: wave w covers co rows [co0+w*16, +16)
-      const ushort* dprow =
-          &dp[(wave * 16 + (lane & 15)) * WGV2_DPW + 8 + kgrp];
-      const s16x8 a_0 = *reinterpret_cast<const s16x8*>(dprow);
-      s16x8 afrag[NTAP == 1 ? 1 : 3];
-      if (KS == 1) {
-        afrag[0] = a_0;
-      } else {
-        const s16x8 a_m = *reinterpret_cast<const s16x8*>(dprow - 8);
-        const s16x8 a_p = *reinterpret_cast<const s16x8*>(dprow + 8);
-        // dp index = u + PAD - kx: kx=0 -> +1, kx=1 -> 0, kx=2 -> -1
-        afrag[0] = __builtin_shufflevector(a_0, a_p, 1, 2, 3, 4, 5, 6, 7, 8);
-        afrag[1] = a_0;
-        afrag[2] =
-            __builtin_shufflevector(a_m, a_0, 7, 8, 9, 10, 11, 12, 13, 14);
-      }
-      if (abl == 2) { __syncthreads(); continue; }
+      if (abl != 2) {
+        // ---- compute: wave w covers co rows [co0+w*16, +16)
 #pragma unroll
-      for (int ky = 0; ky < KS; ++ky) {
-        const int slot = KS == 1 ? oy % 3 : (((oy + ky - PAD) % 3) + 3) % 3;
+        for (int r = 0; r < RB; ++r) {
+          const int oy = oy0 + r;
+          const ushort* dprow = &dp[((r * 64) + wave * 16 + (lane & 15))
+                                    * WGV2_DPW + 8 + kgrp];
+          const s16x8 a_0 = *reinterpret_cast<const s16x8*>(dprow);
+          s16x8 afrag[NTAP == 1 ? 1 : 3];
+          if (KS == 1) {
+            afrag[0] = a_0;
+          } else {
+            const s16x8 a_m = *reinterpret_cast<const s16x8*>(dprow - 8);
+            const s16x8 a_p = *reinterpret_cast<const s16x8*>(dprow + 8);
+            // dp index = u + PAD - kx: kx=0 -> +1, kx=1 -> 0, kx=2 -> -1
+            afrag[0] =
+                __builtin_shufflevector(a_0, a_p, 1, 2, 3, 4, 5, 6, 7, 8);
+            afrag[1] = a_0;
+            afrag[2] =
+                __builtin_shufflevector(a_m, a_0, 7, 8, 9, 10, 11, 12, 13, 14);
+          }
 #pragma unroll
-        for (int nci = 0; nci < 2; ++nci) {
-          const s16x8 bfrag = *reinterpret_cast<const s16x8*>(
-              &xs[((nci * 16 + (lane & 15)) * 3 + slot) * WGV2_XW + kgrp]);
+          for (int ky = 0; ky < KS; ++ky) {
+            const int slot = KS == 1 ? oy % NSLOT
+                : (((oy + ky - PAD) % NSLOT) + NSLOT) % NSLOT;
 #pragma unroll
-          for (int kx = 0; kx < KS; ++kx)
-            acc[ky * KS + kx][nci] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afrag[kx], bfrag, acc[ky * KS + kx][nci], 0, 0, 0);
+            for (int nci = 0; nci < 2; ++nci) {
+              const s16x8 bfrag = *reinterpret_cast<const s16x8*>(
+                  &xs[((nci * 16 + (lane & 15)) * NSLOT + slot) * WGV2_XW
+                      + kgrp]);
+#pragma unroll
+              for (int kx = 0; kx < KS; ++kx)
+                acc[ky * KS + kx][nci] =
+                    __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        afrag[kx], bfrag, acc[ky * KS + kx][nci], 0, 0, 0);
+            }
+          }
         }
       }
       __syncthreads();
@@ -533,8 +547,8 @@ void conv2d_wgrad_s1_kernel(
 
   // ---- flush: plain coalesced stores of this block's partial tile into
   // scratch [g][32ci][64co][NTAP]; a second kernel reduces over g.  No
-  // atomics at all (the atomic flood was 70-87% of this kernel, measured
-  // via ESR_WGRAD_ABL) and the reduction order is DETERMINISTIC.
+  // atomics (an atomic flood here measured 70-87% of the kernel) and the
+  // reduction order is DETERMINISTIC.
   if (abl == 1) return;
   float* part = dwp + ((((long long)blockIdx.x * gridDim.y + blockIdx.y)
                         * gridDim.z + blockIdx.z) * (32 * 64 * NTAP));
