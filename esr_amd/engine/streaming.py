@@ -134,3 +134,21 @@ class StreamingESR:
             except Exception:
                 self.use_graphs = False
         return self._forward(inp)[0]
+
+    @torch.no_grad()
+    def push_events(self, events: torch.Tensor,
+                    capacity: int | None = None,
+                    mode: str = "linear") -> torch.Tensor | None:
+        """Push one LR event window and get the prediction back AS AN HR
+        EVENT STREAM [n, 4] (x, y, t, p sorted by t) — the serving-side
+        count->event conversion (reference cnt2event,
+        ESR:dataloader/cython_cnt2event/cnt2event.pyx:18-116).  On GPU this
+        runs the device-only redistribution pipeline (redistribute.hip);
+        pass `capacity` for a fixed-shape, sync-free conversion."""
+        cnt = self.push(events)
+        if cnt is None:
+            return None
+        ev = E.redistribute_count(cnt.float().round().clamp(min=0)[None],
+                                  mode=mode, capacity=capacity)[0]
+        n = int((ev[:, 3] != 0).sum()) if capacity is None else capacity
+        return ev[:n]

@@ -75,3 +75,46 @@ def test_streaming_graphed_matches_eager_gpu():
     for i in range(2, 6):
         assert torch.allclose(eager[i], graphed[i], atol=1e-4), \
             (i, (eager[i] - graphed[i]).abs().max().item())
+
+
+def test_push_events_roundtrip():
+    """Serving-side count->event output: the emitted HR event stream
+    splats back to the predicted count map."""
+    from esr_amd.engine.streaming import StreamingESR
+    from esr_amd.models import build_model
+    from esr_amd.ops.events import events_to_channels
+    torch.manual_seed(0)
+    model = build_model("ESRNet", inch=2, basech=4, num_frame=3)
+    srv = StreamingESR(model, lr_resolution=(16, 16), scale=2,
+                       device="cpu", use_graphs=False)
+    out_ev = None
+    for i in range(4):
+        g = torch.Generator().manual_seed(i)
+        n = 200
+        ev = torch.stack([
+            torch.randint(0, 16, (n,), generator=g).float(),
+            torch.randint(0, 16, (n,), generator=g).float(),
+            torch.sort(torch.rand(n, generator=g)).values,
+            torch.randint(0, 2, (n,), generator=g).float() * 2 - 1])
+        out_ev = srv.push_events(ev)
+    assert out_ev is not None and out_ev.shape[1] == 4
+    t = out_ev[:, 2]
+    assert (t[1:] >= t[:-1]).all()
+    # splat back == the (rounded, clamped) predicted count map
+    srv2 = StreamingESR(model, lr_resolution=(16, 16), scale=2,
+                        device="cpu", use_graphs=False)
+    model.reset_states()
+    cnt = None
+    for i in range(4):
+        g = torch.Generator().manual_seed(i)
+        n = 200
+        ev = torch.stack([
+            torch.randint(0, 16, (n,), generator=g).float(),
+            torch.randint(0, 16, (n,), generator=g).float(),
+            torch.sort(torch.rand(n, generator=g)).values,
+            torch.randint(0, 2, (n,), generator=g).float() * 2 - 1])
+        cnt = srv2.push(ev)
+    want = cnt.float().round().clamp(min=0)
+    back = events_to_channels(out_ev[:, 0], out_ev[:, 1], out_ev[:, 3],
+                              want.shape[-2:])
+    assert torch.equal(back, want)
