@@ -172,6 +172,8 @@ def conv2d_act(x: torch.Tensor, conv: torch.nn.Conv2d, act: str | None):
     if act not in ACT_IDS or not native_conv_supported(x, conv):
         return None
     if torch.is_autocast_enabled():
+        if torch.get_autocast_dtype("cuda") != torch.bfloat16:
+            return None   # fp16 autocast: kernels are bf16-native; fall back
         x = x.to(torch.bfloat16)
     if x.dtype != torch.bfloat16:
         return None
