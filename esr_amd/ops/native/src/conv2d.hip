@@ -60,7 +60,7 @@ ESR_INLINE ushort f_to_bf16u(float f) {
 // Wave grid 2(M) x 2(N); per wave MREP m-frags x 2 n-frags of 16x16.
 // ---------------------------------------------------------------------------
 
-template <int KS, int STRIDE, int ACT, int MREP, int NCO>
+template <int KS, int STRIDE, int ACT, int MREP>
 __global__ __launch_bounds__(256)
 void conv2d_fwd_mfma_kernel(
     const ushort* __restrict__ x,      // [B, Cin, H, W] bf16
@@ -82,17 +82,14 @@ void conv2d_fwd_mfma_kernel(
   const int ty0 = (blockIdx.x / ntx) * TH;
   const int tx0 = (blockIdx.x % ntx) * TW;
   const int b = blockIdx.y;
-  // NCO cout-blocks per block: the LDS patch (and every B fragment) is
-  // staged once and shared across them — the z>1 shapes (Cout > 64) were
-  // exactly the ones losing to MIOpen before this fold
-  const int co_base = blockIdx.z * (NCO * 32 * MREP);
+  const int co0 = blockIdx.z * (32 * MREP);
 
   const long long x_b = (long long)b * Cin * H * W;
   const long long plane = (long long)H * W;
   const int in_y0 = ty0 * STRIDE - PAD;
   const int in_x0 = tx0 * STRIDE - PAD;
 
-  f32x4 acc[NCO][MREP][2] = {};
+  f32x4 acc[MREP][2] = {};
 
   const int kgrp = (lane >> 4) * 8;  // k-offset of this lane's fragment rows
   const int nchunks = Cin_p / CIK;
@@ -123,22 +120,16 @@ void conv2d_fwd_mfma_kernel(
     }
     __syncthreads();
 
-    // ---- MFMA: taps x NCO x MREP x 2 fragments per wave per chunk
+    // ---- MFMA: 9 (or 1) taps x MREP x 2 fragments per wave per chunk
 #pragma unroll
     for (int tap = 0; tap < KS * KS; ++tap) {
       const int ky = tap / KS, kx = tap % KS;
-      s16x8 a[NCO][MREP];
+      s16x8 a[MREP];
 #pragma unroll
-      for (int cb = 0; cb < NCO; ++cb) {
-#pragma unroll
-        for (int m = 0; m < MREP; ++m) {
-          const int row = co_base + cb * 32 * MREP + wm * 16 * MREP + m * 16
-              + (lane & 15);
-          a[cb][m] = row < Cout_p
-              ? *reinterpret_cast<const s16x8*>(
-                    &wp[((long long)tap * Cout_p + row) * Cin_p + ci0 + kgrp])
-              : s16x8{};
-        }
+      for (int m = 0; m < MREP; ++m) {
+        const int row = co0 + wm * 16 * MREP + m * 16 + (lane & 15);
+        a[m] = *reinterpret_cast<const s16x8*>(
+            &wp[((long long)tap * Cout_p + row) * Cin_p + ci0 + kgrp]);
       }
 #pragma unroll
       for (int nf = 0; nf < 2; ++nf) {
@@ -148,11 +139,9 @@ void conv2d_fwd_mfma_kernel(
         const s16x8 bfrag = *reinterpret_cast<const s16x8*>(
             &patch[(py * PW + px) * SLOT + kgrp]);
 #pragma unroll
-        for (int cb = 0; cb < NCO; ++cb)
-#pragma unroll
-          for (int m = 0; m < MREP; ++m)
-            acc[cb][m][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a[cb][m], bfrag, acc[cb][m][nf], 0, 0, 0);
+        for (int m = 0; m < MREP; ++m)
+          acc[m][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[m], bfrag, acc[m][nf], 0, 0, 0);
       }
     }
   }
@@ -160,24 +149,20 @@ void conv2d_fwd_mfma_kernel(
   // ---- epilogue: bias + activation fused, bf16 stores
   // D layout (guide §3): col = lane&15 (pixel), row = (lane>>4)*4 + j (cout)
 #pragma unroll
-  for (int cb = 0; cb < NCO; ++cb) {
+  for (int m = 0; m < MREP; ++m) {
 #pragma unroll
-    for (int m = 0; m < MREP; ++m) {
+    for (int nf = 0; nf < 2; ++nf) {
+      const int pidx = wn * 32 + nf * 16 + (lane & 15);
+      const int oy = ty0 + (pidx >> 5), ox = tx0 + (pidx & 31);
+      if (oy >= Ho || ox >= Wo) continue;
 #pragma unroll
-      for (int nf = 0; nf < 2; ++nf) {
-        const int pidx = wn * 32 + nf * 16 + (lane & 15);
-        const int oy = ty0 + (pidx >> 5), ox = tx0 + (pidx & 31);
-        if (oy >= Ho || ox >= Wo) continue;
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const int co = co_base + cb * 32 * MREP + wm * 16 * MREP + m * 16
-              + (lane >> 4) * 4 + j;
-          if (co >= Cout) continue;
-          float v = acc[cb][m][nf][j];
-          if (bias) v += bias[co];
-          v = apply_act<ACT>(v);
-          y[(((long long)b * Cout + co) * Ho + oy) * Wo + ox] = f_to_bf16u(v);
-        }
+      for (int j = 0; j < 4; ++j) {
+        const int co = co0 + wm * 16 * MREP + m * 16 + (lane >> 4) * 4 + j;
+        if (co >= Cout) continue;
+        float v = acc[m][nf][j];
+        if (bias) v += bias[co];
+        v = apply_act<ACT>(v);
+        y[(((long long)b * Cout + co) * Ho + oy) * Wo + ox] = f_to_bf16u(v);
       }
     }
   }
@@ -692,29 +677,23 @@ at::Tensor conv2d_fwd_mfma(const at::Tensor& x, const at::Tensor& wp,
     bias_ptr = bias->data_ptr<float>();
   }
   const int mrep = (Cout > 32 && Cout_p % 32 == 0) ? 2 : 1;
-  const int nblk = (Cout_p + 32 * mrep - 1) / (32 * mrep);
-  const int nco = nblk >= 3 ? 3 : nblk;   // cout-blocks folded per block
   const int ntx = (Wo + TW - 1) / TW, nty = (Ho + TH - 1) / TH;
-  dim3 grid(ntx * nty, B, (nblk + nco - 1) / nco);
+  dim3 grid(ntx * nty, B, (Cout_p + 32 * mrep - 1) / (32 * mrep));
   auto stream = at::hip::getCurrentHIPStream();
   DISPATCH_KS_STRIDE((int)ks, (int)stride, [&] {
     DISPATCH_ACT((int)act, [&] {
-      auto launch = [&](auto kern) {
-        hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
-                           (const ushort*)x.data_ptr(),
-                           (const ushort*)wp.data_ptr(),
+      if (mrep == 2)
+        hipLaunchKernelGGL((conv2d_fwd_mfma_kernel<kKS, kST, kAct, 2>),
+                           grid, dim3(256), 0, stream,
+                           (const ushort*)x.data_ptr(), (const ushort*)wp.data_ptr(),
                            bias_ptr, (ushort*)y.data_ptr(),
                            Cin, H, W, (int)Cout, Ho, Wo, Cin_p, Cout_p, ntx);
-      };
-      if (mrep == 2) {
-        if (nco == 3) launch(conv2d_fwd_mfma_kernel<kKS, kST, kAct, 2, 3>);
-        else if (nco == 2) launch(conv2d_fwd_mfma_kernel<kKS, kST, kAct, 2, 2>);
-        else launch(conv2d_fwd_mfma_kernel<kKS, kST, kAct, 2, 1>);
-      } else {
-        if (nco == 3) launch(conv2d_fwd_mfma_kernel<kKS, kST, kAct, 1, 3>);
-        else if (nco == 2) launch(conv2d_fwd_mfma_kernel<kKS, kST, kAct, 1, 2>);
-        else launch(conv2d_fwd_mfma_kernel<kKS, kST, kAct, 1, 1>);
-      }
+      else
+        hipLaunchKernelGGL((conv2d_fwd_mfma_kernel<kKS, kST, kAct, 1>),
+                           grid, dim3(256), 0, stream,
+                           (const ushort*)x.data_ptr(), (const ushort*)wp.data_ptr(),
+                           bias_ptr, (ushort*)y.data_ptr(),
+                           Cin, H, W, (int)Cout, Ho, Wo, Cin_p, Cout_p, ntx);
       return 0;
     });
     return 0;
