@@ -221,3 +221,18 @@ def test_esrnet_forward_uses_native_convs():
     # (Cout>=32 or Cin>=32) must go native
     assert conv_mod.stats["native_calls"] >= 20, \
         f"only {conv_mod.stats['native_calls']} native conv dispatches"
+
+
+def test_wgrad_deterministic_across_runs():
+    """The two-stage wgrad reduction has a FIXED summation order (no fp32
+    atomics): identical inputs give bitwise-identical dW across runs —
+    the reference's atomic col2im cannot promise this (SURVEY §5 race
+    detection)."""
+    from esr_amd.ops.native import require_ext
+    ext = require_ext()
+    x = _rand_bf16(16, 64, 32, 32, seed=3)
+    dy = _rand_bf16(16, 64, 32, 32, seed=4)
+    ref = ext.conv2d_wgrad_mfma(x, dy, 3, 1, 64, 64).clone()
+    for _ in range(3):
+        again = ext.conv2d_wgrad_mfma(x, dy, 3, 1, 64, 64)
+        assert torch.equal(ref, again), "wgrad not bitwise deterministic"
