@@ -42,6 +42,8 @@ at::Tensor conv2d_fwd_mfma(const at::Tensor&, const at::Tensor&,
                            int64_t, int64_t, int64_t, int64_t);
 at::Tensor conv2d_fwd_valu(const at::Tensor&, const at::Tensor&,
                            const c10::optional<at::Tensor>&, int64_t, int64_t);
+at::Tensor conv2d_fwd_valu2(const at::Tensor&, const at::Tensor&,
+                            const c10::optional<at::Tensor>&, int64_t, int64_t);
 at::Tensor conv2d_dgrad_s2(const at::Tensor&, const at::Tensor&,
                            int64_t, int64_t);
 at::Tensor conv2d_wgrad_mfma(const at::Tensor&, const at::Tensor&,
@@ -72,6 +74,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "bf16 NCHW conv fwd, MFMA implicit GEMM, fused bias+act (gfx950)");
   m.def("conv2d_fwd_valu", &conv2d_fwd_valu,
         "bf16 NCHW conv fwd, direct VALU, fused bias+act (gfx950)");
+  m.def("conv2d_fwd_valu2", &conv2d_fwd_valu2,
+        "bf16 NCHW conv fwd, register-strip VALU v2 for tiny channels");
   m.def("conv2d_dgrad_s2", &conv2d_dgrad_s2,
         "bf16 stride-2 conv input-grad (gfx950)");
   m.def("conv2d_wgrad_mfma", &conv2d_wgrad_mfma,

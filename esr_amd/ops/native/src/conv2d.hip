@@ -213,6 +213,122 @@ __global__ void conv2d_fwd_valu_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// VALU forward v2 for the tiny-channel shapes (head 2->8, enc1 8->16,
+// tail 8->2 at 256x256): each thread computes an 8-wide output strip for
+// up to 8 cout at once — input row segments live in registers and are
+// reused across cout and taps, weights broadcast from LDS.  The v1
+// per-output kernel was instruction-bound (one bounds-checked scalar load
+// per MAC); this one does ~18 loads per 1152 MACs on the head shape.
+// ---------------------------------------------------------------------------
+
+template <int KS, int STRIDE, int ACT, int COCH>
+__global__ __launch_bounds__(256)
+void conv2d_fwd_valu2_kernel(
+    const ushort* __restrict__ x,   // [B, Cin, H, W] bf16
+    const ushort* __restrict__ w,   // [Cout, Cin, KS, KS] bf16
+    const float* __restrict__ bias, ushort* __restrict__ y,
+    int Cin, int H, int W, int Cout, int Ho, int Wo,
+    long long nstrips, int nsx) {
+  constexpr int PAD = KS / 2;
+  constexpr int VEC = 8;                       // outputs per thread strip
+  constexpr int NPX = VEC * STRIDE + KS - 1;   // input px per row segment
+  __shared__ ushort wl[COCH * 32 * KS * KS];   // this co-chunk x <=32ci
+
+  // ---- preload this co-chunk's weights once per block
+  const int co0 = blockIdx.z * COCH;
+  const int nw = COCH * Cin * KS * KS;
+  for (int i = threadIdx.x; i < nw; i += 256) {
+    const int co = i / (Cin * KS * KS);
+    wl[i] = (co0 + co < Cout)
+        ? w[((long long)(co0 + co) * Cin) * KS * KS + (i % (Cin * KS * KS))]
+        : (ushort)0;
+  }
+  __syncthreads();
+
+  for (long long s = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       s < nstrips; s += (long long)gridDim.x * blockDim.x) {
+    const int sx = (int)(s % nsx);
+    const int oy = (int)((s / nsx) % Ho);
+    const int b = (int)(s / ((long long)nsx * Ho));
+    const int ox0 = sx * VEC;
+    const long long plane = (long long)H * W;
+    const ushort* xb = x + (long long)b * Cin * plane;
+
+    float acc[COCH][VEC];   // compile-time indexed everywhere (rule 20)
+#pragma unroll
+    for (int c = 0; c < COCH; ++c)
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) acc[c][j] = 0.f;
+
+    const int ix0 = ox0 * STRIDE - PAD;
+    const bool interior = ix0 >= 0 && ix0 + NPX <= W;
+    for (int ci = 0; ci < Cin; ++ci) {
+      const ushort* xp = xb + ci * plane;
+#pragma unroll
+      for (int ky = 0; ky < KS; ++ky) {
+        const int iy = oy * STRIDE + ky - PAD;
+        float seg[NPX];
+        if (iy >= 0 && iy < H) {
+          const ushort* row = xp + (long long)iy * W;
+          if (interior) {
+#pragma unroll
+            for (int e = 0; e < NPX; ++e)
+              seg[e] = bf16u_to_f(row[ix0 + e]);
+          } else {
+#pragma unroll
+            for (int e = 0; e < NPX; ++e) {
+              const int ix = ix0 + e;
+              seg[e] = (ix >= 0 && ix < W) ? bf16u_to_f(row[ix]) : 0.f;
+            }
+          }
+        } else {
+#pragma unroll
+          for (int e = 0; e < NPX; ++e) seg[e] = 0.f;
+        }
+        const ushort* wrow = &wl[(ci * KS + ky) * KS];
+#pragma unroll
+        for (int c = 0; c < COCH; ++c) {
+          float wk[KS];
+#pragma unroll
+          for (int kx = 0; kx < KS; ++kx)
+            wk[kx] = bf16u_to_f(wrow[c * Cin * KS * KS + kx]);
+#pragma unroll
+          for (int j = 0; j < VEC; ++j) {
+            float v = acc[c][j];
+#pragma unroll
+            for (int kx = 0; kx < KS; ++kx)
+              v = fmaf(seg[j * STRIDE + kx], wk[kx], v);
+            acc[c][j] = v;
+          }
+        }
+      }
+    }
+
+    // ---- epilogue: bias + act, vector bf16 stores per cout row
+    const int nvalid = Wo - ox0 < VEC ? Wo - ox0 : VEC;
+#pragma unroll
+    for (int c = 0; c < COCH; ++c) {
+      const int co = co0 + c;
+      if (co >= Cout) break;
+      ushort packed[VEC];
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float v = acc[c][j];
+        if (bias) v += bias[co];
+        packed[j] = f_to_bf16u(apply_act<ACT>(v));
+      }
+      ushort* dst = y + (((long long)b * Cout + co) * Ho + oy) * Wo + ox0;
+      if (nvalid == VEC) {
+        *reinterpret_cast<s16x8*>(dst) =
+            *reinterpret_cast<const s16x8*>(packed);
+      } else {
+        for (int j = 0; j < nvalid; ++j) dst[j] = packed[j];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Stride-2 input-grad (scatter form as a gather): dx[iy][ix] sums the <=4
 // taps whose (iy + PAD - ky) is even, over all Cout.
 // ---------------------------------------------------------------------------
@@ -723,6 +839,49 @@ at::Tensor conv2d_fwd_valu(const at::Tensor& x, const at::Tensor& w,
                          dim3(esr_grid(n)), dim3(ESR_BLOCK), 0, stream,
                          n, (const ushort*)x.data_ptr(), (const ushort*)w.data_ptr(),
                          bias_ptr, (ushort*)y.data_ptr(), Cin, H, W, Cout, Ho, Wo);
+      return 0;
+    });
+    return 0;
+  });
+  C10_HIP_KERNEL_LAUNCH_CHECK();
+  return y;
+}
+
+// v2 strip kernel for tiny-channel shapes (see kernel comment)
+at::Tensor conv2d_fwd_valu2(const at::Tensor& x, const at::Tensor& w,
+                            const c10::optional<at::Tensor>& bias,
+                            int64_t stride, int64_t act) {
+  check_bf16_4d(x, "conv2d_fwd_valu2: x");
+  check_bf16_4d(w, "conv2d_fwd_valu2: w");
+  const int B = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int Cout = w.size(0), ks = w.size(2);
+  TORCH_CHECK(Cin <= 32, "valu2 is for tiny-channel shapes (Cin <= 32)");
+  const int Ho = (H + 2 * (ks / 2) - ks) / (int)stride + 1;
+  const int Wo = (W + 2 * (ks / 2) - ks) / (int)stride + 1;
+  auto y = at::empty({B, Cout, Ho, Wo}, x.options());
+  const float* bias_ptr = nullptr;
+  if (bias.has_value()) bias_ptr = bias->data_ptr<float>();
+  const int coch = Cout >= 8 || Cout > 2 ? 8 : 2;
+  const int nsx = (Wo + 7) / 8;
+  const long long nstrips = (long long)B * Ho * nsx;
+  dim3 grid(esr_grid(nstrips), 1, (Cout + coch - 1) / coch);
+  auto stream = at::hip::getCurrentHIPStream();
+  DISPATCH_KS_STRIDE(ks, (int)stride, [&] {
+    DISPATCH_ACT((int)act, [&] {
+      if (coch == 8)
+        hipLaunchKernelGGL((conv2d_fwd_valu2_kernel<kKS, kST, kAct, 8>),
+                           grid, dim3(256), 0, stream,
+                           (const ushort*)x.data_ptr(),
+                           (const ushort*)w.data_ptr(), bias_ptr,
+                           (ushort*)y.data_ptr(), Cin, H, W, Cout, Ho, Wo,
+                           nstrips, nsx);
+      else
+        hipLaunchKernelGGL((conv2d_fwd_valu2_kernel<kKS, kST, kAct, 2>),
+                           grid, dim3(256), 0, stream,
+                           (const ushort*)x.data_ptr(),
+                           (const ushort*)w.data_ptr(), bias_ptr,
+                           (ushort*)y.data_ptr(), Cin, H, W, Cout, Ho, Wo,
+                           nstrips, nsx);
       return 0;
     });
     return 0;

@@ -51,6 +51,46 @@ def bench(fn, iters, warmup=10):
     return (time.perf_counter() - t0) / iters * 1e3
 
 
+def valu2_component(args):
+    """Tiny-channel shapes: v2 strip kernel vs v1 VALU vs torch/MIOpen."""
+    from esr_amd.ops.conv import ACT_IDS
+    from esr_amd.ops.native import require_ext
+    ext = require_ext()
+    dev = "cuda:0"
+    torch.manual_seed(0)
+    tiny = [sh for sh in SHAPES if sh[2] < 32 and sh[3] < 32]
+    print(f"{'shape':34s} {'v2 ms':>9s} {'v1 ms':>9s} {'torch':>9s} {'t/v2':>6s}")
+    for label, B, cin, cout, h, w, ks, stride, act in tiny:
+        x = torch.randn(B, cin, h, w, device=dev).to(torch.bfloat16)
+        wt = (torch.randn(cout, cin, ks, ks, device=dev) * 0.2) \
+            .to(torch.bfloat16)
+        b = torch.randn(cout, device=dev).float()
+        aid = ACT_IDS[act]
+        act_fn = {None: lambda t: t, "relu": F.relu,
+                  "sigmoid": torch.sigmoid, "tanh": torch.tanh}[act]
+
+        def v2():
+            return ext.conv2d_fwd_valu2(x, wt, b, stride, aid)
+
+        def v1():
+            return ext.conv2d_fwd_valu(x, wt, b, stride, aid)
+
+        def ref():
+            return act_fn(F.conv2d(x, wt, b.to(torch.bfloat16),
+                                   stride=stride, padding=ks // 2))
+        # correctness spot check before timing
+        got = v2().float()
+        want = act_fn(F.conv2d(x.float(), wt.float(), b,
+                               stride=stride, padding=ks // 2))
+        err = (got - want).abs().max().item()
+        scale = want.abs().max().item() + 1e-6
+        assert err / scale < 2e-2, f"{label}: v2 wrong, rel {err / scale}"
+        t2 = bench(v2, args.iters)
+        t1 = bench(v1, args.iters)
+        tt = bench(ref, args.iters)
+        print(f"{label:34s} {t2:9.3f} {t1:9.3f} {tt:9.3f} {tt / t2:6.2f}")
+
+
 def wgrad_component(args):
     """Isolated weight-grad: native conv2d_wgrad_mfma vs aten wgrad-only."""
     from esr_amd.ops.native import require_ext
@@ -89,10 +129,14 @@ def main():
     ap.add_argument("--bwd", action="store_true", help="also time fwd+bwd")
     ap.add_argument("--wgrad", action="store_true",
                     help="time the wgrad kernel alone vs aten wgrad-only")
+    ap.add_argument("--valu2", action="store_true",
+                    help="A/B the tiny-channel strip kernel vs v1 and torch")
     args = ap.parse_args()
 
     if args.wgrad:
         return wgrad_component(args)
+    if args.valu2:
+        return valu2_component(args)
 
     from esr_amd.ops.conv import ACT_IDS, _NativeConv2dFn
     dev = "cuda:0"
