@@ -88,6 +88,11 @@ class _NativeConv2dFn(torch.autograd.Function):
         if cout >= _MFMA_MIN_COUT:
             y = ext.conv2d_fwd_mfma(x, _pack(w), bias_f, cout, ks, stride,
                                     act_id)
+        elif cin < _MFMA_MIN_COUT:
+            # tiny-channel head/tail class: register-strip kernel
+            # (2.7x / 1.3x MIOpen on head/tail, bench_conv --valu2)
+            y = ext.conv2d_fwd_valu2(x, w.contiguous(), bias_f, stride,
+                                     act_id)
         else:
             y = ext.conv2d_fwd_valu(x, w.contiguous(), bias_f, stride, act_id)
         ctx.save_for_backward(x, w, y)
@@ -148,9 +153,11 @@ def native_conv_supported(x: torch.Tensor, conv: torch.nn.Conv2d) -> bool:
     if not (_enabled() and x.is_cuda and get_ext() is not None):
         return False
     if conv.out_channels < _MFMA_MIN_COUT and conv.in_channels < _MFMA_MIN_COUT:
-        # head/enc1/tail class: MIOpen measured faster than the v1 VALU
-        # kernel on these bandwidth-bound shapes (tools/bench_conv.py)
-        return False
+        # tiny-channel class: the v2 register-strip kernel beats MIOpen at
+        # stride 1 (head 2.7x, tail 1.3x); the stride-2 enc1 shape still
+        # loses (0.67x) and stays on MIOpen (tools/bench_conv.py --valu2)
+        if conv.stride[0] != 1:
+            return False
     if conv.groups != 1 or conv.dilation != (1, 1):
         return False
     kh, kw = conv.kernel_size
