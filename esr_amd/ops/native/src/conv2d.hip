@@ -318,7 +318,9 @@ void conv2d_fwd_valu2_kernel(
         packed[j] = f_to_bf16u(apply_act<ACT>(v));
       }
       ushort* dst = y + (((long long)b * Cout + co) * Ho + oy) * Wo + ox0;
-      if (nvalid == VEC) {
+      // vector store only when the row base is 16B-aligned (Wo % 8 != 0
+      // leaves interior strips misaligned)
+      if (nvalid == VEC && (((unsigned long long)(uintptr_t)dst) & 15) == 0) {
         *reinterpret_cast<s16x8*>(dst) =
             *reinterpret_cast<const s16x8*>(packed);
       } else {
