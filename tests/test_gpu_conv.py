@@ -76,7 +76,8 @@ CASES = [
     (40, 48, 30, 30, 3, 1, "tanh"),      # unaligned channels + odd spatial
     (16, 33, 20, 44, 3, 1, None),        # Cout one past tile edge
     (64, 128, 32, 32, 3, 1, None),       # pixel-shuffle pre-conv
-    (8, 2, 62, 30, 3, 1, "relu"),        # tail class, odd W (valu2 edges)
+    (8, 2, 62, 32, 3, 1, "relu"),        # tail class, COCH=2, odd H
+    (8, 2, 62, 30, 3, 1, "relu"),        # odd W -> routed to the v1 kernel
     (2, 8, 64, 64, 3, 2, None),          # tiny s2 (valu2 strided segs)
 ]
 
