@@ -91,9 +91,11 @@ class _NativeConv2dFn(torch.autograd.Function):
         elif cin < _MFMA_MIN_COUT and x.shape[-1] % 8 == 0:
             # tiny-channel head/tail class: register-strip kernel
             # (2.7x / 1.3x MIOpen on head/tail, bench_conv --valu2).
-            # Widths not divisible by 8 have a known v2 defect (odd-W
-            # oracle case) and take the always-correct v1 kernel below;
-            # every real model shape is /8-padded (ESRNet.DOWN_SCALE).
+            # Widths not divisible by 8 conservatively take the v1 kernel
+            # (their strip stores are row-misaligned; the suspected odd-W
+            # failure later turned out to be a test-harness fp32-mean
+            # artifact, but v2 at odd W is GPU-unverified) — every real
+            # model shape is /8-padded anyway (ESRNet.DOWN_SCALE).
             y = ext.conv2d_fwd_valu2(x, w.contiguous(), bias_f, stride,
                                      act_id)
         else:
