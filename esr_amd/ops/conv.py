@@ -11,10 +11,10 @@ MI355X, profiles/README.md):
     bf16 on the deep/mid shapes;
   * direct VALU kernel for Cin >= 32 with tiny Cout (attention/kernel
     convs, 1.2-1.3x);
-  * shapes with BOTH Cin < 32 and Cout < 32 (head/enc1/tail) fall back to
-    torch/MIOpen — the v1 VALU kernel measured 0.2-0.7x there (scalar
-    loads + per-tap branches; an instruction-count-optimized v2 is the
-    open item).
+  * register-strip VALU v2 for the tiny-channel class (Cin and Cout both
+    < 32): 8-wide output strips x all cout per thread, weights broadcast
+    from LDS — head 2.7x, tail 1.3x vs MIOpen; the stride-2 enc1 shape
+    (0.67x) stays on MIOpen.
 Backward defaults to torch's aten.convolution_backward (MIOpen bf16).
 The native backward kernels are implemented, oracle-tested and measured
 (tools/bench_conv.py --wgrad / --bwd, MI355X):
