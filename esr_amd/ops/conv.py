@@ -154,9 +154,8 @@ class _NativeConv2dFn(torch.autograd.Function):
         return dx, dw, db, None, None
 
 
-def native_conv_supported(x: torch.Tensor, conv: torch.nn.Conv2d) -> bool:
-    if not (_enabled() and x.is_cuda and get_ext() is not None):
-        return False
+def shape_supported(conv: torch.nn.Conv2d) -> bool:
+    """Shape half of the dispatch predicate (CPU-testable)."""
     if conv.out_channels < _MFMA_MIN_COUT and conv.in_channels < _MFMA_MIN_COUT:
         # tiny-channel class: the v2 register-strip kernel beats MIOpen at
         # stride 1 (head 2.7x, tail 1.3x); the stride-2 enc1 shape still
@@ -176,6 +175,12 @@ def native_conv_supported(x: torch.Tensor, conv: torch.nn.Conv2d) -> bool:
     if ph != kh // 2:
         return False
     return True
+
+
+def native_conv_supported(x: torch.Tensor, conv: torch.nn.Conv2d) -> bool:
+    if not (_enabled() and x.is_cuda and get_ext() is not None):
+        return False
+    return shape_supported(conv)
 
 
 def conv2d_act(x: torch.Tensor, conv: torch.nn.Conv2d, act: str | None):
