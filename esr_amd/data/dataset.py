@@ -93,6 +93,14 @@ class EventSRDataset(Dataset):
         self.add_noise = config.get("add_noise", {"enabled": False})
         self.augment_cfg = config.get("data_augment", {"enabled": False})
         self.hot_filter = config.get("hot_filter", {"enabled": False})
+        # selective item fields: the flagship trainer consumes 3 of the 21
+        # item-dict entries; listing them here skips the unused encodings
+        # and interpolations in the CPU workers (~3x items/s at the
+        # flagship shape — the 8-GPU input-feed budget, docs/SCALING.md).
+        # None (default) builds everything (reference item-dict parity,
+        # ESR:dataloader/h5dataset.py:374-406).
+        f = config.get("fields", None)
+        self.fields = None if f is None else set(f)
 
         (self.inp_prefix, self.inp_sensor_resolution, self.gt_prefix,
          self.gt_sensor_resolution, self.inp_down_sensor_resolution) = \
@@ -294,65 +302,100 @@ class EventSRDataset(Dataset):
         inp_res = self.inp_sensor_resolution
         gt_res = self.gt_sensor_resolution
 
-        inp_stack = E.events_to_stack_no_polarity(inp[0], inp[1], inp[2], inp[3],
-                                                  self.time_bins, inp_res)
-        inp_cnt = E.events_to_channels(inp[0], inp[1], inp[3], inp_res)
+        fields = self.fields
+        custom_on = self.custom_resolution is not None
 
-        if self.hot_filter.get("enabled", False):
+        def need(*keys):
+            return fields is None or any(k in fields for k in keys)
+
+        hot_on = self.hot_filter.get("enabled", False)
+        need_cnt = hot_on or custom_on or \
+            need("inp_cnt", "inp_bicubic_cnt", "inp_near_cnt")
+        need_stack = hot_on or \
+            need("inp_stack", "inp_bicubic_stack", "inp_near_stack")
+        need_norm = custom_on or \
+            need("inp_scaled_cnt", "inp_scaled_stack",
+                 "inp_down_cnt", "inp_down_scaled_cnt")
+        need_down = custom_on or need("inp_down_cnt", "inp_down_scaled_cnt")
+
+        out = {"gt_img": gt_img, "gt_inp_size_img": gt_img_inp,
+               "frame": frame}
+        inp_cnt = inp_stack = None
+        if need_stack:
+            inp_stack = E.events_to_stack_no_polarity(
+                inp[0], inp[1], inp[2], inp[3], self.time_bins, inp_res)
+        if need_cnt:
+            inp_cnt = E.events_to_channels(inp[0], inp[1], inp[3], inp_res)
+        if hot_on:
             hot_mask = self._hot_mask(inp, inp_res)
             inp_cnt = inp_cnt * hot_mask
             inp_stack = inp_stack * hot_mask
-        inp_bicubic_cnt = self._interp(inp_cnt, gt_res, "bicubic")
-        inp_bicubic_stack = self._interp(inp_stack, gt_res, "bicubic")
-        inp_near_cnt = self._interp(inp_cnt, gt_res, "nearest")
-        inp_near_stack = self._interp(inp_stack, gt_res, "nearest")
+        if inp_cnt is not None:
+            out["inp_cnt"] = inp_cnt
+        if inp_stack is not None:
+            out["inp_stack"] = inp_stack
+        if need("inp_bicubic_cnt"):
+            out["inp_bicubic_cnt"] = self._interp(inp_cnt, gt_res, "bicubic")
+        if need("inp_bicubic_stack"):
+            out["inp_bicubic_stack"] = self._interp(inp_stack, gt_res,
+                                                    "bicubic")
+        if need("inp_near_cnt"):
+            out["inp_near_cnt"] = self._interp(inp_cnt, gt_res, "nearest")
+        if need("inp_near_stack"):
+            out["inp_near_stack"] = self._interp(inp_stack, gt_res, "nearest")
 
-        norm_ev = E.normalize_events(inp, inp_res)
-        inp_scaled_cnt = E.scaled_count_encoding(norm_ev, gt_res, "cnt")
-        inp_scaled_stack = E.scaled_count_encoding(norm_ev, gt_res, "stack",
-                                                   self.time_bins)
-        inp_down_cnt, inp_down_scaled_cnt = self._unsupervised_pair(norm_ev)
+        inp_scaled_cnt = inp_down_cnt = inp_down_scaled_cnt = None
+        if need_norm:
+            norm_ev = E.normalize_events(inp, inp_res)
+            if custom_on or need("inp_scaled_cnt"):
+                inp_scaled_cnt = E.scaled_count_encoding(norm_ev, gt_res,
+                                                         "cnt")
+                out["inp_scaled_cnt"] = inp_scaled_cnt
+            if need("inp_scaled_stack"):
+                out["inp_scaled_stack"] = E.scaled_count_encoding(
+                    norm_ev, gt_res, "stack", self.time_bins)
+            if need_down:
+                inp_down_cnt, inp_down_scaled_cnt = \
+                    self._unsupervised_pair(norm_ev)
+                out["inp_down_cnt"] = inp_down_cnt
+                out["inp_down_scaled_cnt"] = inp_down_scaled_cnt
 
-        gt_stack = E.events_to_stack_no_polarity(gt[0], gt[1], gt[2], gt[3],
-                                                 self.time_bins, gt_res)
-        gt_cnt = E.events_to_channels(gt[0], gt[1], gt[3], gt_res)
+        if need("gt_stack"):
+            out["gt_stack"] = E.events_to_stack_no_polarity(
+                gt[0], gt[1], gt[2], gt[3], self.time_bins, gt_res)
+        gt_cnt = None
+        if custom_on or need("gt_cnt"):
+            gt_cnt = E.events_to_channels(gt[0], gt[1], gt[3], gt_res)
+            out["gt_cnt"] = gt_cnt
 
-        if self.custom_resolution is not None:
+        if custom_on:
             cr = list(self.custom_resolution)
             cr_up = [c * self.scale for c in cr]
             cr_dn = [round(c / self.scale) for c in cr]
-            custom = [
-                self._interp(inp_cnt, cr, "bicubic").round(),
-                self._interp(inp_scaled_cnt, cr_up, "bicubic").round(),
-                self._interp(inp_down_cnt, cr_dn, "bicubic").round(),
-                self._interp(inp_down_scaled_cnt, cr, "bicubic").round(),
-                self._interp(gt_cnt, cr_up, "bicubic").round(),
-            ]
-        else:
-            custom = [torch.zeros_like(inp_cnt) for _ in range(5)]
+            out["inp_custom_cnt"] = self._interp(inp_cnt, cr,
+                                                 "bicubic").round()
+            out["inp_custom_scaled_cnt"] = self._interp(
+                inp_scaled_cnt, cr_up, "bicubic").round()
+            out["inp_custom_down_cnt"] = self._interp(
+                inp_down_cnt, cr_dn, "bicubic").round()
+            out["inp_custom_down_scaled_cnt"] = self._interp(
+                inp_down_scaled_cnt, cr, "bicubic").round()
+            out["gt_custom_cnt"] = self._interp(gt_cnt, cr_up,
+                                                "bicubic").round()
+        elif fields is None:
+            zero = torch.zeros_like(
+                inp_cnt if inp_cnt is not None
+                else torch.zeros([2] + list(inp_res)))
+            for k in ("inp_custom_cnt", "inp_custom_scaled_cnt",
+                      "inp_custom_down_cnt", "inp_custom_down_scaled_cnt",
+                      "gt_custom_cnt"):
+                out[k] = zero.clone()
 
-        return {
-            "inp_stack": inp_stack,
-            "inp_cnt": inp_cnt,
-            "inp_bicubic_cnt": inp_bicubic_cnt,
-            "inp_bicubic_stack": inp_bicubic_stack,
-            "inp_near_cnt": inp_near_cnt,
-            "inp_near_stack": inp_near_stack,
-            "inp_scaled_cnt": inp_scaled_cnt,
-            "inp_scaled_stack": inp_scaled_stack,
-            "inp_down_cnt": inp_down_cnt,
-            "inp_down_scaled_cnt": inp_down_scaled_cnt,
-            "inp_custom_cnt": custom[0],
-            "inp_custom_scaled_cnt": custom[1],
-            "inp_custom_down_cnt": custom[2],
-            "inp_custom_down_scaled_cnt": custom[3],
-            "gt_custom_cnt": custom[4],
-            "gt_stack": gt_stack,
-            "gt_cnt": gt_cnt,
-            "gt_img": gt_img,
-            "gt_inp_size_img": gt_img_inp,
-            "frame": frame,
-        }
+        if fields is not None:
+            return {k: v for k, v in out.items()
+                    if k in fields or k in ("gt_img", "gt_inp_size_img",
+                                            "frame")}
+        return out
 
     def _unsupervised_pair(self, norm_ev):
         """Down-scaled self-supervision pair (parity:

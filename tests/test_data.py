@@ -224,3 +224,19 @@ def test_redistribute_all_empty():
     from esr_amd.ops import redistribute_stack
     out = redistribute_stack(torch.zeros(3, 4, 5, 5))
     assert out.shape == (3, 1, 4) and out.abs().sum() == 0
+
+
+def test_selective_fields(store_path):
+    """dataset.fields restricts the item dict to the requested encodings
+    (plus the always-present frame slots) and the values match the
+    full-dict ones bit for bit."""
+    full = EventSRDataset(store_path, _ds_config())
+    lean = EventSRDataset(store_path, _ds_config(
+        fields=["inp_scaled_cnt", "gt_cnt", "inp_cnt"]))
+    a = full.__getitem__(1, seed=7)
+    b = lean.__getitem__(1, seed=7)
+    assert set(b) == {"inp_scaled_cnt", "gt_cnt", "inp_cnt",
+                      "gt_img", "gt_inp_size_img", "frame"}
+    for k in ("inp_scaled_cnt", "gt_cnt", "inp_cnt"):
+        assert torch.equal(a[k], b[k]), k
+    assert len(a) > len(b)

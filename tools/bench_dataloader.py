@@ -44,6 +44,8 @@ def main():
     ap.add_argument("--seconds", type=float, default=12.0)
     ap.add_argument("--workers", type=int, nargs="*", default=[0, 2, 4, 8])
     ap.add_argument("--batch", type=int, default=8)
+    ap.add_argument("--fields", action="store_true",
+                    help="restrict items to the trainer's 3 fields")
     args = ap.parse_args()
 
     from esr_amd.data import SequenceDataLoader, make_synthetic_dataset
@@ -55,7 +57,10 @@ def main():
     print(f"{'workers':>8s} {'items/s':>10s} {'seq items/s':>12s}")
     results = {}
     for w in args.workers:
-        dl = SequenceDataLoader(_dl_config(datalist, w, args.batch))
+        cfg = _dl_config(datalist, w, args.batch)
+        if args.fields:
+            cfg["dataset"]["fields"] = ["inp_scaled_cnt", "gt_cnt", "inp_cnt"]
+        dl = SequenceDataLoader(cfg)
         it = iter(dl)
         next(it)  # warm workers
         n = 0
