@@ -60,12 +60,18 @@ def main():
     p.add_argument("--real_world_test", action="store_true")
     p.add_argument("--no_images", action="store_true")
     p.add_argument("--num_workers", type=int, default=2)
-    p.add_argument("--lpips_weights", type=str, default=None)
+    p.add_argument("--lpips_weights", type=str, default=None,
+                   help="linear-head state dict (defaults to the bundled "
+                        "reference v0.1 heads)")
+    p.add_argument("--lpips_backbone", type=str, default=None,
+                   help="torchvision alexnet/vgg16 checkpoint for "
+                        "paper-comparable LPIPS")
     args = p.parse_args()
 
     device = torch.device(args.device if torch.cuda.is_available() else "cpu")
     dl_cfg = default_dataloader_config(args)
-    metrics = build_metrics(device, lpips_weights=args.lpips_weights)
+    metrics = build_metrics(device, lpips_weights=args.lpips_weights,
+                            lpips_backbone=args.lpips_backbone)
 
     if args.model_list:
         model_paths = read_datalist(args.model_list)
