@@ -99,7 +99,7 @@ class FrameInterpolator(nn.Module):
         self.flow_net = InterpUNet(2, 4, base)          # grayscale pair -> flows
         self.refine_net = InterpUNet(8, 5, base)  # I0,I1,g0,g1,Ft0,Ft1
 
-    def forward(self, I0, I1, t: float):
+    def forward(self, I0, I1, t: float, return_flows: bool = False):
         flows = self.flow_net(torch.cat([I0, I1], dim=1))
         F01 = flows[:, 0:2]
         F10 = flows[:, 2:4]
@@ -117,7 +117,10 @@ class FrameInterpolator(nn.Module):
         g1 = backwarp(I1, Ft1)
         num = (1 - t) * V0 * g0 + t * V1 * g1
         den = (1 - t) * V0 + t * V1
-        return num / (den + 1e-8)
+        out = num / (den + 1e-8)
+        if return_flows:  # for self-training (tools/train_interp.py)
+            return out, (F01, F10, Ft0, Ft1)
+        return out
 
 
 @torch.no_grad()
