@@ -268,10 +268,12 @@ def redistribute_stack(stack: torch.Tensor, mode: str = "linear",
         order = torch.argsort(ev[:, 2], stable=True)
         per_item.append(ev[order])
 
-    maxlen = max(e.size(0) for e in per_item)
+    maxlen = max(e.size(0) for e in per_item) if capacity is None \
+        else int(capacity)
     out = torch.zeros(Bb, maxlen, 4, device=device)
     for b, e in enumerate(per_item):
-        out[b, : e.size(0)] = e
+        n = min(e.size(0), maxlen)   # fixed capacity: truncate like the
+        out[b, :n] = e[:n]           # HIP pipeline (cell-order prefix)
     return out
 
 

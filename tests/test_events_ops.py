@@ -196,3 +196,20 @@ def test_scaled_encoding_stack_and_events_modes():
     evm = scaled_count_encoding(norm, (8, 8), "events")
     assert evm.shape == (4, 5)
     assert evm[0, 0] == 2 and evm[1, 2] == 6
+
+
+def test_redistribute_capacity_cpu():
+    """The torch path honours an explicit capacity like the HIP pipeline:
+    fixed output shape, zero padding, truncation when over."""
+    from esr_amd.ops.events import redistribute_stack
+    g = torch.Generator().manual_seed(4)
+    stack = torch.randint(0, 3, (2, 2, 8, 8), generator=g).float()
+    n0 = int(stack[0].abs().sum())
+    full = redistribute_stack(stack, mode="linear")
+    padded = redistribute_stack(stack, mode="linear", capacity=n0 + 16)
+    assert padded.shape[1] == n0 + 16
+    assert torch.equal(padded[:, : full.shape[1]], full)
+    assert (padded[:, full.shape[1]:] == 0).all()
+    small = redistribute_stack(stack, mode="linear", capacity=5)
+    assert small.shape[1] == 5
+    assert torch.equal(small[0], full[0, :5])
