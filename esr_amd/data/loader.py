@@ -123,6 +123,18 @@ def SequenceDataLoader(dataloader_config):
     loader.gt_sensor_resolution = ds.datasets[0].gt_sensor_resolution
     loader.inp_sensor_resolution = ds.datasets[0].inp_sensor_resolution
     loader.scale = dataloader_config["dataset"]["scale"]
+
+    def set_epoch(epoch: int):
+        """Advance the per-index augmentation seeds (and the sharded
+        sampler's shuffle) — item randomness is epoch-keyed, not
+        worker-keyed, so runs reproduce for any num_workers.
+        NOTE: with persistent_workers the new epoch takes effect at the
+        next iterator creation (workers re-pickle the dataset)."""
+        for d in ds.datasets:
+            d.set_epoch(epoch)
+        if sampler is not None:
+            sampler.set_epoch(epoch)
+    loader.set_epoch = set_epoch
     return loader
 
 

@@ -272,7 +272,10 @@ class Trainer:
         epoch = 0
         done = False
         while not done:
-            if self.train_dataloader.dist_sampler is not None:
+            if hasattr(self.train_dataloader, "set_epoch"):
+                # epoch-keyed item seeds + sharded-sampler shuffle
+                self.train_dataloader.set_epoch(epoch)
+            elif self.train_dataloader.dist_sampler is not None:
                 self.train_dataloader.dist_sampler.set_epoch(epoch)
             for idx, inputs_seq in enumerate(self.train_dataloader):
                 iter_idx = idx + len(self.train_dataloader) * epoch \
@@ -333,7 +336,9 @@ class Trainer:
 
     def epoch_based_training(self):
         for epoch in range(self.start_epoch, self.epochs + 1):
-            if self.train_dataloader.dist_sampler is not None:
+            if hasattr(self.train_dataloader, "set_epoch"):
+                self.train_dataloader.set_epoch(epoch)
+            elif self.train_dataloader.dist_sampler is not None:
                 self.train_dataloader.dist_sampler.set_epoch(epoch)
             self.model.train()
             self.train_metrics.reset()
