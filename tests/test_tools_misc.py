@@ -36,3 +36,26 @@ def test_baseline_config_arms_spec():
     # config 5 is the fp16 DVS-native arm
     f5 = bbc.ARMS["config5_dvs_fp16"]["flags"]
     assert "fp16" in f5 and "180" in f5 and "240" in f5
+
+
+def test_txt_to_evs_cli(tmp_path):
+    import numpy as np
+    rng = np.random.default_rng(0)
+    n = 500
+    txt = tmp_path / "ev.txt"
+    data = np.stack([np.sort(rng.uniform(0, 1, n)),
+                     rng.integers(0, 64, n),
+                     rng.integers(0, 48, n),
+                     rng.integers(0, 2, n)], axis=1)
+    np.savetxt(txt, data)
+    out = subprocess.run(
+        [sys.executable, str(REPO / "tools" / "txt_to_evs.py"), str(txt),
+         str(tmp_path / "seq.evs"), "--height", "48", "--width", "64"],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr
+    sys.path.insert(0, str(REPO))
+    from esr_amd.data.store import EventStore
+    st = EventStore(tmp_path / "seq.evs")
+    assert st.num_events("ori") == n
+    ev = st.events("down2", 0, st.num_events("down2"))
+    assert ev[0].max() < 32 and ev[1].max() < 24
