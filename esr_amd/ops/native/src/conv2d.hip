@@ -18,8 +18,10 @@
 //
 // Backward: stride-1 input-grad IS this forward with flipped/transposed
 // packed weights (host side); stride-2 input-grad has a dedicated
-// scatter-form VALU kernel; weight-grad is the MFMA split-K kernel below
-// (fp32 atomics into a padded dW).
+// scatter-form VALU kernel; stride-1 weight-grad is the MFMA kernel with
+// a DETERMINISTIC two-stage reduction (conv2d_wgrad_s1_kernel below),
+// stride-2 weight-grad the older per-row-tile variant with transposed
+// line-parallel atomics.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -372,13 +374,13 @@ __global__ void conv2d_dgrad_s2_valu_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// MFMA weight-grad, split-K over output pixels with fp32 atomics.
+// MFMA weight-grad v1 (kept for STRIDE 2), split-K over output pixels.
 // dW[co][ci][ky][kx] = sum_pix dpre[co][oy][ox] * X[ci][oy*s+ky-1][ox*s+kx-1]
-// reindexed over u = ox*s + kx - PAD so the X tile stays aligned:
 //   A = dpre, pre-shifted per kx (3 small scalar-staged copies, halo'd)
 //   B = X, one aligned vector-staged copy [ci][ky-row][u-chunk]
-// Each wave owns its own LDS tiles and an independent slice of output rows
-// (no cross-wave barriers); one atomicAdd per output element at the end.
+// Each wave owns its own LDS tiles and an independent slice of output
+// rows; the flush reduces the 4 waves through LDS then issues atomics in
+// the TRANSPOSED [Cin_p][Cout_p] scratch (lanes span cachelines).
 // ---------------------------------------------------------------------------
 
 constexpr int WG_PW = 40;  // padded 32-px rows, same bank math as SLOT
@@ -388,7 +390,7 @@ __global__ __launch_bounds__(256)
 void conv2d_wgrad_mfma_kernel(
     const ushort* __restrict__ x,     // [B, Cin, H, W]
     const ushort* __restrict__ dpre,  // [B, Cout, Ho, Wo]
-    float* __restrict__ dwp,          // [Cout_p, Cin_p, KS, KS] fp32 (zeroed)
+    float* __restrict__ dwp,          // TRANSPOSED [Cin_p, Cout_p, KS, KS]
     int Cin, int H, int W, int Cout, int Ho, int Wo,
     int Cin_p, int Cout_p, int rows_per_blk) {
   constexpr int PAD = KS / 2;
