@@ -39,11 +39,20 @@ def barrier():
 
 
 def init_distributed(backend: str | None = None) -> int:
-    """Initialize the process group from torchrun env vars.
+    """Initialize the process group from torchrun env vars (SLURM launches
+    supported via SLURM_PROCID, parity: ESR:train_ours_cnt_seq.py:64-85).
 
     Returns the local device index.  Single-process (no RANK in env) runs
     skip initialization and return 0 — the framework works unlaunched.
     """
+    if "RANK" not in os.environ and "SLURM_PROCID" in os.environ \
+            and "WORLD_SIZE" in os.environ:
+        # srun launch: derive rank/local rank from the SLURM env
+        os.environ["RANK"] = os.environ["SLURM_PROCID"]
+        os.environ.setdefault(
+            "LOCAL_RANK",
+            str(int(os.environ["SLURM_PROCID"]) %
+                max(torch.cuda.device_count(), 1)))
     if "RANK" not in os.environ or "WORLD_SIZE" not in os.environ:
         return 0
     rank = int(os.environ["RANK"])
